@@ -1,0 +1,266 @@
+"""The ACCO decoupled-training engine: the two-round state machine.
+
+Reimplements (from scratch, MI355X-first) the algorithm of reference
+trainer_decoupled.py:
+
+- `gradient_step`            (:18-39)   — micro-batch forward/backward into
+                                          the flat grad arena,
+- `update_buffers_step`      (:43-63)   — params ⇄ com-buffer swap,
+- `communication_round`      (:67-126)  — count all-reduce + bucketed
+                                          reduce-scatter + sharded AdamW +
+                                          bucketed all-gather,
+- `com_routine` / `train_acco` (:129-168, :431-520) — the compute∥comm
+                                          thread choreography on two HIP
+                                          streams,
+- `warmup_steps`             (:318-383) — sequential warmup rounds,
+- `train_dpu`                (:605-663) — the single-stream DPU baseline.
+
+ACCO semantics (must hold exactly — SURVEY.md §3.1): gradients accumulate
+across two com rounds and are zeroed only after even rounds; the even round
+takes a *tentative* optimizer step (commit=False — no state mutation,
+replacing the reference's snapshot/rollback) to predict the next params so
+compute never stalls; the odd round takes the true step on the full
+two-half-round gradient; the scheduler and the global grad counter advance
+only on odd rounds, by the globally-summed grad count (which is what makes
+heterogeneous per-rank accumulation correct).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Callable, Dict, Optional
+
+import torch
+
+from acco_amd.engine.sharded_adamw import ShardedAdamW
+from acco_amd.engine.scheduler import LRSchedule
+from acco_amd.engine.streams import Event, Stream
+from acco_amd.parallel.comm import CommBackend, ShardSpec
+
+
+class AccoEngine:
+    """Owns the flat buffers and runs ACCO / DPU rounds.
+
+    Parameters
+    ----------
+    params_arena : flat (bf16) parameter vector, model params alias it.
+    grads_arena  : flat grad vector of the same padded size (spec.total).
+    spec         : bucket-major shard geometry.
+    comm         : collective backend (RCCL / gloo).
+    forward_backward : callable(inputs) -> loss float tensor; must
+        accumulate gradients into grads_arena (i.e. call loss.backward()).
+    """
+
+    def __init__(self, *, params_arena: torch.Tensor, grads_arena: torch.Tensor,
+                 n_live: int, spec: ShardSpec, comm: CommBackend, rank: int,
+                 device: torch.device, opt: ShardedAdamW, sched: LRSchedule,
+                 forward_backward: Callable[[Dict], torch.Tensor],
+                 next_batch: Callable[[], Dict],
+                 n_grad_accumulation: int = 1,
+                 log=None):
+        self.params = params_arena
+        self.grads = grads_arena
+        self.n = n_live
+        self.spec = spec
+        self.comm = comm
+        self.rank = rank
+        self.device = device
+        self.opt = opt
+        self.sched = sched
+        self.forward_backward = forward_backward
+        self.next_batch = next_batch
+        self.n_acc = n_grad_accumulation
+        self.log = log
+
+        # the communication buffer (reference com_buffer, :244-269)
+        self.com_buffer = torch.zeros(spec.total, dtype=params_arena.dtype,
+                                      device=device)
+        int_t = torch.int32
+        self.count_grad_local = torch.zeros(1, dtype=int_t, device=device)
+        self.count_grad_this_round = torch.zeros(1, dtype=int_t, device=device)
+        self.loss_static = torch.zeros(1, dtype=torch.float32, device=device)
+
+        self.com_stream = Stream(device)
+        self.grad_stream = Stream(device)
+        self.com_event = Event(device)
+        self.update_event = Event(device)
+        self.end_of_grad = Event(device)
+
+        self.round_idx = 0            # reference count_after_init
+        self.count_grad_tot = 0
+        self.com_finished = threading.Event()
+
+        # hooks the trainer installs (logging / eval / checkpoint); called
+        # on the compute thread after each buffer update, rank 0 only.
+        self.on_round_complete: Optional[Callable[[int, int], None]] = None
+        # test instrumentation: when set to a list, records the per-round
+        # LOCAL grad count at each buffer update (oracle replay support)
+        self.trace: Optional[list] = None
+
+    # ------------------------------------------------------------ pieces
+
+    def gradient_step(self, inputs: Dict) -> None:
+        """One micro-batch forward/backward accumulating into grads arena
+        (reference gradient_step :18-39)."""
+        loss = self.forward_backward(inputs)
+        self.count_grad_local += 1
+        self.loss_static.copy_(loss.detach().float().reshape(1))
+
+    @torch.no_grad()
+    def update_buffers_step(self, zero_grads: bool) -> None:
+        """Swap: params ← buffer (new params); buffer ← accumulated grads;
+        publish local grad count; optionally zero grads+count
+        (reference update_buffers_step :43-63)."""
+        self.params[:self.n].copy_(self.com_buffer[:self.n])
+        self.com_buffer[:self.n].copy_(self.grads[:self.n])
+        if self.com_buffer.numel() > self.n:
+            self.com_buffer[self.n:].zero_()
+        self.count_grad_this_round.copy_(self.count_grad_local)
+        if self.trace is not None:
+            self.trace.append(int(self.count_grad_local.item()))
+        if zero_grads:
+            self.grads.zero_()
+            self.count_grad_local.zero_()
+
+    @torch.no_grad()
+    def communication_round(self, commit: bool, advance_sched: bool = None) -> int:
+        """One com round on the com buffer: C2 count all-reduce (async) →
+        per-bucket C3 reduce-scatter → fused sharded AdamW per bucket as its
+        bucket lands (pipelined) → per-bucket C4 all-gather
+        (reference communication_step :67-126). Returns the global grad
+        count of the round."""
+        if advance_sched is None:
+            advance_sched = commit
+        work_count = self.comm.all_reduce_sum_async(self.count_grad_this_round)
+        rs = [self.comm.reduce_scatter_bucket_async(self.com_buffer, self.spec,
+                                                    j, self.rank)
+              for j in range(self.spec.nb)]
+        work_count.wait()
+        # grad averaging divides by the GLOBAL grad count (reference :85-98)
+        inv_count = 1.0 / self.count_grad_this_round.float()
+        lr = self.sched.lr()
+        ag = []
+        for j in range(self.spec.nb):
+            rs[j].wait()
+            self.opt.step_bucket(j, self.com_buffer, grad_scale=inv_count,
+                                 commit=commit, lr=lr)
+            ag.append(self.comm.all_gather_bucket_async(self.com_buffer,
+                                                        self.spec, j, self.rank))
+        for w in ag:
+            w.wait()
+        self.opt.finish_round(commit)
+        n_global = int(self.count_grad_this_round.item())
+        if advance_sched:
+            self.sched.advance(n_global)
+        return n_global
+
+    # ------------------------------------------------------- init / warmup
+
+    def bootstrap(self, n_warmup_steps: int) -> None:
+        """Fill the com buffer so the first threaded round has gradients
+        (reference prepare_buffer_com :262-267 + warmup_steps :318-383 +
+        train_acco init :436-442)."""
+        if n_warmup_steps > 0:
+            # buffer starts holding the params; warmup rounds run sequentially
+            self.com_buffer[:self.n].copy_(self.params[:self.n])
+            for _ in range(n_warmup_steps):
+                self.params[:self.n].copy_(self.com_buffer[:self.n])
+                for _ in range(self.n_acc):
+                    self.gradient_step(self.next_batch())
+                self.update_buffers_step(zero_grads=True)
+                self.communication_round(commit=True)
+            # tail: one more grad round so the threaded loop starts with
+            # fresh grads in the buffer (reference :358-383)
+            self.params[:self.n].copy_(self.com_buffer[:self.n])
+            for _ in range(self.n_acc):
+                self.gradient_step(self.next_batch())
+            self.update_buffers_step(zero_grads=True)
+            self.count_grad_tot = self.comm.world * (n_warmup_steps + 1) * self.n_acc
+        else:
+            # bootstrap gradient: one real forward/backward seeds the buffer
+            # (reference prepare_grads + init_count=1, :262-267,441)
+            self.gradient_step(self.next_batch())
+            self.com_buffer[:self.n].copy_(self.grads[:self.n])
+            self.count_grad_this_round.fill_(1)
+            # grads stay un-zeroed: round 0's compute accumulates onto them
+            # and the tentative round consumes the buffered copy (reference
+            # semantics — count_grad_local continues from 1).
+            self.count_grad_tot = 0
+        self.round_idx = 0
+
+    # --------------------------------------------------------- ACCO loop
+
+    def train_acco(self, nb_grad_tot: int, n_warmup_steps: int = 0) -> None:
+        self.bootstrap(n_warmup_steps)
+        barrier = threading.Barrier(2)
+        stop = threading.Event()
+
+        def com_routine():
+            try:
+                self.com_stream.wait_default(self.device)
+                with self.com_stream.activate():
+                    while self.count_grad_tot < nb_grad_tot and not stop.is_set():
+                        commit = (self.round_idx % 2 == 1)
+                        self.communication_round(commit=commit)
+                        self.com_event.record_and_sync(self.com_stream)
+                        self.com_finished.set()
+                        barrier.wait()
+            except threading.BrokenBarrierError:
+                pass
+            except Exception:
+                # a dead com thread must not leave the compute thread parked
+                barrier.abort()
+                raise
+
+        com_thread = threading.Thread(target=com_routine, daemon=True)
+        com_thread.start()
+
+        try:
+            self.grad_stream.wait_default(self.device)
+            with self.grad_stream.activate():
+                while self.count_grad_tot < nb_grad_tot:
+                    for _ in range(self.n_acc):
+                        self.gradient_step(self.next_batch())
+                    self.end_of_grad.record_and_sync(self.grad_stream)
+                    if self.com_finished.is_set():
+                        self.com_finished.clear()
+                        if self.round_idx % 2 == 1:
+                            self.count_grad_tot += int(
+                                self.count_grad_this_round.item())
+                        self.update_buffers_step(
+                            zero_grads=(self.round_idx % 2 == 0))
+                        self.update_event.record_and_sync(self.grad_stream)
+                        self.round_idx += 1
+                        barrier.wait()
+                        if self.on_round_complete is not None and self.rank == 0:
+                            self.on_round_complete(self.round_idx,
+                                                   self.count_grad_tot)
+        finally:
+            stop.set()
+            # unblock the com thread if it is parked at the barrier
+            try:
+                barrier.abort()
+            except Exception:
+                pass
+            com_thread.join(timeout=60.0)
+
+    # ---------------------------------------------------------- DPU loop
+
+    def train_dpu(self, nb_grad_tot: int, n_warmup_steps: int = 0) -> None:
+        """Delayed-parameter-update baseline: same primitives, sequential,
+        stale gradients, full step every round (reference train_dpu
+        :605-663 — communication_step there always runs with the default
+        count_after_init=-1, i.e. commit+schedule every round, while
+        update_buffers_step alternates grad zeroing)."""
+        self.bootstrap(n_warmup_steps)
+        count_com = 0
+        while self.count_grad_tot < nb_grad_tot:
+            for _ in range(self.n_acc):
+                self.gradient_step(self.next_batch())
+            self.communication_round(commit=True)
+            self.count_grad_tot += int(self.count_grad_this_round.item())
+            self.update_buffers_step(zero_grads=(count_com % 2 == 0))
+            count_com += 1
+            if self.on_round_complete is not None and self.rank == 0:
+                self.on_round_complete(count_com, self.count_grad_tot)

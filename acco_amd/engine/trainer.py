@@ -1,0 +1,359 @@
+"""DecoupledTrainer — the public API (signature parity with the reference).
+
+``DecoupledTrainer(model, tokenizer, train_dataset, eval_dataset, args, log,
+text_column_name, preprocess_dataset_fn, run_name)`` + ``.train()``
+(reference trainer_decoupled.py:170-197,418-429; README.md:100-109), built
+on the MI355X-native engine: flat bf16 arenas, bucket-major RCCL
+collectives over xGMI, fused gfx950 sharded AdamW, and the ACCO / DDP /
+DPU training modes selected by ``args.method_name``.
+"""
+
+from __future__ import annotations
+
+import contextlib
+import os
+import time
+from typing import Dict, Optional
+
+import torch
+from torch.utils.data import DataLoader, Dataset, RandomSampler
+
+from acco_amd.data.packing import (make_tokenize_const_len_fn,
+                                   make_tokenize_truncate_fn)
+from acco_amd.data.synthetic import collate_input_ids
+from acco_amd.engine import arena
+from acco_amd.engine.acco import AccoEngine
+from acco_amd.engine.bootstrap import DistContext, init_distributed
+from acco_amd.engine.scheduler import LRSchedule
+from acco_amd.engine.sharded_adamw import ShardedAdamW
+from acco_amd.parallel.comm import CommBackend, ShardSpec
+from acco_amd.parallel.ddp import NativeZeroDDP
+from acco_amd.utils.logging import (ScalarLogger, create_dict_result,
+                                    get_logger, print_training_evolution,
+                                    save_result)
+
+
+class DecoupledTrainer:
+    def __init__(self, model=None, tokenizer=None, train_dataset=None,
+                 eval_dataset=None, args=None, log=None,
+                 text_column_name="text", preprocess_dataset_fn=None,
+                 run_name=""):
+        self.args = args
+        self.tokenizer = tokenizer
+        self.text_column_name = text_column_name
+        self.log = log or get_logger()
+        self.run_name = run_name
+        self.nb_grad_tot = int(args.nb_steps_tot)
+        self.epoch = 0
+
+        # ---- distributed bootstrap (reference initialize_com :135-180)
+        self.ctx: DistContext = init_distributed()
+        self.rank = self.ctx.rank
+        self.local_rank = self.ctx.local_rank
+        self.world_size = self.ctx.world_size
+        self.n_nodes = self.ctx.n_nodes
+        self.device = self.ctx.device
+        self.id_run = self.ctx.id_run
+
+        self.dtype = torch.bfloat16 if args.use_mixed_precision else torch.float32
+        if self.device.type != "cuda" and self.dtype == torch.bfloat16:
+            # CPU test path: bf16 CPU training is slow and noisy; keep the
+            # arena dtype but run matmuls in fp32 via autocast off.
+            pass
+
+        # ---- flat arenas; spec first so both arenas are padded to B
+        n_live = arena.live_numel(model)
+        self.spec = ShardSpec.build(n_live, self.world_size,
+                                    buckets=int(getattr(args, "comm_buckets", 8) or 8))
+        self.model = model
+        self.params = arena.flatten_params(model, self.dtype, self.device,
+                                           pad_to=self.spec.total)
+        self.grads = arena.attach_grad_arena(model, self.dtype, self.device,
+                                             pad_to=self.spec.total)
+        self.n_live = n_live
+
+        self.comm = CommBackend(self.device)
+        # C1: average random-init weights across ranks (reference :180)
+        self.comm.all_reduce_avg(self.params)
+
+        # ---- observability
+        out_dir = os.getcwd()
+        self.scalars = ScalarLogger(os.path.join(out_dir, "scalars"),
+                                    run_name or "run", self.id_run, self.rank)
+        self.loss_div = 1.0
+
+        # ---- data
+        self._prepare_data(train_dataset, eval_dataset, preprocess_dataset_fn)
+        self.train_dataloader = self._make_dataloader(self.train_dataset,
+                                                      shuffle=True)
+        self.eval_dataloader = (self._make_dataloader(self.eval_dataset,
+                                                      shuffle=False)
+                                if self.eval_dataset is not None else None)
+        self.train_iterator = iter(self.train_dataloader)
+
+        # ---- optimizer / scheduler / engine
+        self.sched = LRSchedule(args.learning_rate, args.warmup,
+                                self.nb_grad_tot, args.scheduler_name)
+        self.opt = ShardedAdamW(self.spec, self.rank, self.device,
+                                lr=args.learning_rate,
+                                betas=(args.adam_beta1, args.adam_beta2),
+                                eps=float(getattr(args, "adam_eps", 1e-8) or 1e-8),
+                                weight_decay=args.weight_decay)
+
+        if args.method_name == "ddp" or getattr(args, "run_baseline_ddp", False):
+            self.loss_div = float(args.n_grad_accumulation)
+            self.opt.init_master_from_buffer(self.params)
+            self.ddp = NativeZeroDDP(model, self.params, self.grads,
+                                     self.n_live, self.spec, self.comm,
+                                     self.rank, self.opt)
+            self.engine = None
+        else:
+            self.engine = AccoEngine(
+                params_arena=self.params, grads_arena=self.grads,
+                n_live=self.n_live, spec=self.spec, comm=self.comm,
+                rank=self.rank, device=self.device, opt=self.opt,
+                sched=self.sched,
+                forward_backward=self.forward_backward,
+                next_batch=self.load_next_batch,
+                n_grad_accumulation=int(args.n_grad_accumulation),
+                log=self.log)
+            # seed fp32 master from current (averaged) params
+            self.opt.init_master_from_buffer(self.params)
+            self.engine.on_round_complete = self._on_round_complete
+            self.ddp = None
+
+        self.t_beg = time.time()
+        self.t_last_epoch = self.t_beg
+        self._time_checkpoint = time.time()
+        self._last_eval = 0
+
+    # ------------------------------------------------------------- data
+
+    def _prepare_data(self, train_dataset, eval_dataset, preprocess_fn):
+        """Shard by rank and tokenize/pack HF datasets
+        (reference prepare_data :183-200 + tokenize maps :100-125)."""
+        self.train_dataset = train_dataset
+        self.eval_dataset = eval_dataset
+        if train_dataset is None:
+            return
+        is_hf = hasattr(train_dataset, "column_names")
+        if is_hf:
+            self.train_dataset = train_dataset.shard(
+                num_shards=self.world_size, index=self.rank)
+            if eval_dataset is not None:
+                self.eval_dataset = eval_dataset.shard(
+                    num_shards=self.world_size, index=self.rank)
+            if preprocess_fn is not None:
+                self.train_dataset = self.train_dataset.map(preprocess_fn,
+                                                            batched=True)
+                if self.eval_dataset is not None:
+                    self.eval_dataset = self.eval_dataset.map(preprocess_fn,
+                                                              batched=True)
+            if "input_ids" not in self.train_dataset.column_names:
+                if self.args.const_len_batch:
+                    fn = make_tokenize_const_len_fn(
+                        self.tokenizer, self.text_column_name,
+                        self.args.max_length)
+                else:
+                    fn = make_tokenize_truncate_fn(
+                        self.tokenizer, self.text_column_name,
+                        self.args.max_length)
+                cols = self.train_dataset.column_names
+                self.train_dataset = self.train_dataset.map(
+                    fn, batched=True, remove_columns=cols)
+                if self.eval_dataset is not None:
+                    self.eval_dataset = self.eval_dataset.map(
+                        fn, batched=True,
+                        remove_columns=self.eval_dataset.column_names)
+        # torch Datasets (synthetic) are assumed already per-rank seeded
+
+    def _make_dataloader(self, dataset, shuffle: bool) -> DataLoader:
+        kwargs = dict(
+            batch_size=self.args.batch_size,
+            collate_fn=collate_input_ids,
+            num_workers=int(self.args.dataloader_num_workers or 0),
+            drop_last=True,
+        )
+        if self.device.type == "cuda":
+            kwargs["pin_memory"] = bool(self.args.dataloader_pin_memory)
+            kwargs["pin_memory_device"] = f"cuda:{self.device.index}"
+        if kwargs["num_workers"] > 0:
+            kwargs["persistent_workers"] = bool(
+                self.args.dataloader_persistent_workers)
+        if shuffle:
+            kwargs["sampler"] = RandomSampler(dataset)
+        return DataLoader(dataset, **kwargs)
+
+    def load_next_batch(self) -> Dict:
+        """(reference load_next_batch_into_static_memory :386-397)"""
+        try:
+            inputs = next(self.train_iterator)
+        except StopIteration:
+            self.train_iterator = iter(self.train_dataloader)
+            inputs = next(self.train_iterator)
+        return {k: v.to(device=self.device, non_blocking=True)
+                for k, v in inputs.items()}
+
+    # ---------------------------------------------------------- compute
+
+    def _autocast(self):
+        if self.device.type == "cuda" and self.args.use_mixed_precision:
+            return torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
+    def forward_backward(self, inputs: Dict) -> torch.Tensor:
+        """(reference gradient_step :18-39; loss divided by n_acc only in
+        DDP mode — ACCO averages by global grad count instead)"""
+        with self._autocast():
+            if "labels" in inputs:
+                outputs = self.model(**inputs)
+            else:
+                outputs = self.model(**inputs, labels=inputs["input_ids"])
+            loss = outputs[0] / self.loss_div
+        loss.backward()
+        return loss.detach() * self.loss_div
+
+    @torch.no_grad()
+    def eval_loop(self) -> float:
+        """(reference eval_loop :399-415)"""
+        self.model.eval()
+        losses = []
+        for inputs in self.eval_dataloader:
+            inputs = {k: v.to(device=self.device) for k, v in inputs.items()}
+            with self._autocast():
+                if "labels" in inputs:
+                    out = self.model(**inputs)
+                else:
+                    out = self.model(**inputs, labels=inputs["input_ids"])
+            losses.append(out[0].float().cpu())
+        self.model.train()
+        mean = float(torch.stack(losses).mean()) if losses else float("nan")
+        self.log.info(f"eval_loss={mean:.4f}")
+        return mean
+
+    # ------------------------------------------------------ logging hook
+
+    def _on_round_complete(self, round_idx: int, count_grad_tot: int) -> None:
+        eval_loss = None
+        if self.args.eval and self.eval_dataloader is not None:
+            if count_grad_tot - self._last_eval > self.args.eval_step:
+                eval_loss = self.eval_loop()
+                self._last_eval = count_grad_tot
+        loss = float(self.engine.loss_static.item()) if self.engine else 0.0
+        self.scalars.log_training(round_idx // 2, count_grad_tot, self.rank,
+                                  loss, eval_loss, self.t_beg)
+        self.epoch, self.t_last_epoch = print_training_evolution(
+            self.log, count_grad_tot, round_idx, 10, self.rank, self.t_beg,
+            self.t_last_epoch, loss, self.epoch)
+        if self.args.save:
+            if time.time() - self._time_checkpoint >= 1800:
+                self._time_checkpoint = time.time()
+                self.save_model_checkpoint(suffix=f"_model_{count_grad_tot}")
+
+    # ------------------------------------------------------- checkpoints
+
+    def save_model_checkpoint(self, suffix: str = "_model") -> str:
+        """HF-layout model state_dict (reference :559-574)."""
+        path_dir = os.path.join(os.getcwd(), "checkpoints")
+        os.makedirs(path_dir, exist_ok=True)
+        path = os.path.join(path_dir, f"{self.id_run}{suffix}.pt")
+        torch.save(self.model.state_dict(), path)
+        return path
+
+    def save_checkpoint(self, path: str) -> None:
+        """Full resume checkpoint (model + optimizer shard + scheduler +
+        round counters) — a capability the reference lacks (save-only,
+        SURVEY.md §5)."""
+        torch.save({
+            "model": self.model.state_dict(),
+            "opt": self.opt.state_dict(),
+            "sched": self.sched.state_dict(),
+            "round_idx": self.engine.round_idx if self.engine else 0,
+            "count_grad_tot": (self.engine.count_grad_tot
+                               if self.engine else 0),
+        }, path)
+
+    def load_checkpoint(self, path: str) -> None:
+        sd = torch.load(path, map_location=self.device, weights_only=False)
+        self.model.load_state_dict(sd["model"])
+        # refresh the flat arena view values (params alias arena already)
+        self.opt.load_state_dict(sd["opt"])
+        self.sched.load_state_dict(sd["sched"])
+        if self.engine:
+            self.engine.round_idx = int(sd["round_idx"])
+            self.engine.count_grad_tot = int(sd["count_grad_tot"])
+
+    # ------------------------------------------------------------ train
+
+    def train(self) -> None:
+        """(reference train :418-429)"""
+        method = self.args.method_name
+        if method == "acco":
+            self.engine.train_acco(self.nb_grad_tot,
+                                   int(self.args.n_warmup_steps))
+        elif method == "dpu":
+            self.engine.train_dpu(self.nb_grad_tot,
+                                  int(self.args.n_warmup_steps))
+        elif method == "ddp":
+            self.train_ddp()
+        else:
+            raise ValueError(
+                "You must select one of the following method_name: "
+                "'acco', 'ddp', 'dpu'")
+        self._finalize()
+
+    def train_ddp(self) -> None:
+        """Synchronous baseline loop (reference train_ddp :732-763) on the
+        native ZeRO-DDP (bucketed reduce-scatter overlapped with backward)."""
+        count_grad_tot = 0
+        count_com = 0
+        n_acc = int(self.args.n_grad_accumulation)
+        loss_val = 0.0
+        while count_grad_tot < self.nb_grad_tot:
+            for step in range(n_acc):
+                if step == n_acc - 1:
+                    self.ddp.begin_sync_microbatch()
+                inputs = self.load_next_batch()
+                loss = self.forward_backward(inputs)
+                loss_val = float(loss.float().item())
+            lr = self.sched.lr()
+            self.ddp.finish_step(grad_scale=1.0 / self.world_size, lr=lr)
+            self.ddp.zero_grad()
+            self.sched.advance(self.world_size * n_acc)
+            count_grad_tot += self.world_size * n_acc
+            count_com += 1
+            if self.rank == 0:
+                eval_loss = None
+                if self.args.eval and self.eval_dataloader is not None:
+                    if count_grad_tot - self._last_eval > self.args.eval_step:
+                        eval_loss = self.eval_loop()
+                        self._last_eval = count_grad_tot
+                self.scalars.log_training(count_com, count_grad_tot,
+                                          self.rank, loss_val, eval_loss,
+                                          self.t_beg)
+                self.epoch, self.t_last_epoch = print_training_evolution(
+                    self.log, count_grad_tot, count_com, 10, self.rank,
+                    self.t_beg, self.t_last_epoch, loss_val, self.epoch)
+                if self.args.save and time.time() - self._time_checkpoint >= 1800:
+                    self._time_checkpoint = time.time()
+                    self.save_model_checkpoint(
+                        suffix=f"_ddp_model_{count_grad_tot}")
+
+    def _finalize(self) -> None:
+        total_time = time.time() - self.t_beg
+        if self.rank == 0:
+            loss = (float(self.engine.loss_static.item())
+                    if self.engine else 0.0)
+            try:
+                dict_args = self.args.to_container()
+            except AttributeError:
+                dict_args = dict(self.args)
+            device_name = (torch.cuda.get_device_name()
+                           if self.device.type == "cuda" else "cpu")
+            row = create_dict_result(dict_args, self.world_size, self.n_nodes,
+                                     device_name, total_time, self.id_run,
+                                     loss)
+            save_result(os.path.join(os.getcwd(), "results.csv"), row)
+            if self.args.save:
+                self.save_model_checkpoint()
+        self.scalars.close()

@@ -1,0 +1,136 @@
+"""GPT-Neo-family causal LM, MI355X-native.
+
+From-scratch module with ``state_dict`` keys matching HF
+``GPTNeoForCausalLM`` (transformer.wte/wpe, transformer.h.N.ln_1,
+attn.attention.{q,k,v,out}_proj, mlp.c_fc/c_proj, transformer.ln_f,
+lm_head tied to wte), so HF checkpoints load directly.
+
+GPT-Neo specifics faithfully kept (reference config/model/gpt-neo-125M.json):
+- learned absolute position embeddings (wpe);
+- alternating global / local (banded, window 256) causal attention;
+- NO attention scaling (scale = 1.0 — GPT-Neo quirk);
+- gelu_new MLP activation; q/k/v projections have no bias, out_proj does.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from acco_amd import ops
+from acco_amd.models.config import GPTNeoConfig
+
+
+class GPTNeoSelfAttention(nn.Module):
+    def __init__(self, cfg: GPTNeoConfig, attention_type: str):
+        super().__init__()
+        d = cfg.hidden_size
+        self.cfg = cfg
+        self.attention_type = attention_type
+        self.k_proj = nn.Linear(d, d, bias=False)
+        self.v_proj = nn.Linear(d, d, bias=False)
+        self.q_proj = nn.Linear(d, d, bias=False)
+        self.out_proj = nn.Linear(d, d, bias=True)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, S, d = x.shape
+        H, hd = self.cfg.num_heads, self.cfg.head_dim
+        q = self.q_proj(x).view(B, S, H, hd).transpose(1, 2)
+        k = self.k_proj(x).view(B, S, H, hd).transpose(1, 2)
+        v = self.v_proj(x).view(B, S, H, hd).transpose(1, 2)
+        window = self.cfg.window_size if self.attention_type == "local" else None
+        o = ops.causal_attention(q, k, v, scale=1.0, window=window)
+        o = o.transpose(1, 2).reshape(B, S, d)
+        return self.out_proj(o)
+
+
+class GPTNeoAttention(nn.Module):
+    """Wrapper level kept so state_dict keys read attn.attention.* like HF."""
+
+    def __init__(self, cfg: GPTNeoConfig, layer_id: int):
+        super().__init__()
+        self.attention = GPTNeoSelfAttention(cfg, cfg.layer_attention_type(layer_id))
+
+    def forward(self, x):
+        return self.attention(x)
+
+
+class GPTNeoMLP(nn.Module):
+    def __init__(self, cfg: GPTNeoConfig):
+        super().__init__()
+        self.c_fc = nn.Linear(cfg.hidden_size, cfg.inner_size, bias=True)
+        self.c_proj = nn.Linear(cfg.inner_size, cfg.hidden_size, bias=True)
+
+    def forward(self, x):
+        return self.c_proj(ops.gelu_new(self.c_fc(x)))
+
+
+class GPTNeoBlock(nn.Module):
+    def __init__(self, cfg: GPTNeoConfig, layer_id: int):
+        super().__init__()
+        self.ln_1 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
+        self.attn = GPTNeoAttention(cfg, layer_id)
+        self.ln_2 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
+        self.mlp = GPTNeoMLP(cfg)
+
+    def forward(self, x):
+        x = x + self.attn(ops.layer_norm(x, self.ln_1.weight, self.ln_1.bias,
+                                         self.ln_1.eps))
+        x = x + self.mlp(ops.layer_norm(x, self.ln_2.weight, self.ln_2.bias,
+                                        self.ln_2.eps))
+        return x
+
+
+class GPTNeoModel(nn.Module):
+    def __init__(self, cfg: GPTNeoConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.wte = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.wpe = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
+        self.h = nn.ModuleList(GPTNeoBlock(cfg, i) for i in range(cfg.num_layers))
+        self.ln_f = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        S = input_ids.shape[1]
+        pos = torch.arange(S, device=input_ids.device)
+        x = self.wte(input_ids) + self.wpe(pos)[None]
+        for block in self.h:
+            x = block(x)
+        return ops.layer_norm(x, self.ln_f.weight, self.ln_f.bias, self.ln_f.eps)
+
+
+class GPTNeoForCausalLM(nn.Module):
+    config_class = GPTNeoConfig
+
+    def __init__(self, cfg: GPTNeoConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.transformer = GPTNeoModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.transformer.wte.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.cfg.initializer_range
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(0.0, std)
+            if module.bias is not None:
+                module.bias.data.zero_()
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(0.0, std)
+        elif isinstance(module, nn.LayerNorm):
+            module.weight.data.fill_(1.0)
+            module.bias.data.zero_()
+
+    def forward(self, input_ids: torch.Tensor,
+                labels: Optional[torch.Tensor] = None,
+                attention_mask: Optional[torch.Tensor] = None):
+        hidden = self.transformer(input_ids)
+        logits = self.lm_head(hidden)
+        if labels is None:
+            return (logits,)
+        loss = ops.causal_lm_loss(logits, labels)
+        return (loss, logits)

@@ -1,0 +1,143 @@
+"""Llama-family causal LM, MI355X-native.
+
+A from-scratch module whose ``state_dict`` keys match HF
+``LlamaForCausalLM`` (model.embed_tokens / model.layers.N.self_attn.q_proj /
+mlp.gate_proj / input_layernorm / model.norm / lm_head), so HF checkpoints
+load directly (checkpoint-layout parity with the reference, which saves HF
+``state_dict`` — reference trainer_decoupled.py:559-574).
+
+All hot ops route through acco_amd.ops: RMSNorm, RoPE, SwiGLU, causal flash
+attention and the CE loss are hand-written gfx950 HIP kernels on GPU
+(SURVEY.md §2.5 K1/K2 rebuild); GEMMs go to hipBLASLt via torch.nn.Linear.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from acco_amd import ops
+from acco_amd.models.config import LlamaConfig
+
+
+class LlamaRMSNorm(nn.Module):
+    def __init__(self, hidden_size: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.variance_epsilon = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rms_norm(x, self.weight, self.variance_epsilon)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        d, hd = cfg.hidden_size, cfg.head_dim
+        self.q_proj = nn.Linear(d, cfg.num_heads * hd, bias=False)
+        self.k_proj = nn.Linear(d, cfg.num_kv_heads * hd, bias=False)
+        self.v_proj = nn.Linear(d, cfg.num_kv_heads * hd, bias=False)
+        self.o_proj = nn.Linear(cfg.num_heads * hd, d, bias=False)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        B, S, _ = x.shape
+        cfg = self.cfg
+        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim).transpose(1, 2)
+        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2)
+        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2)
+        q, k = ops.rope_apply(q, k, cos, sin)
+        o = ops.causal_attention(q, k, v)          # [B, H, S, D]
+        o = o.transpose(1, 2).reshape(B, S, -1)
+        return self.o_proj(o)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg)
+        self.mlp = LlamaMLP(cfg)
+        self.input_layernorm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.post_attention_layernorm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+
+    def forward(self, x, cos, sin):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(LlamaDecoderLayer(cfg) for _ in range(cfg.num_layers))
+        self.norm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        # RoPE tables are buffers recomputed lazily per device/seq-len
+        # (host-precomputed trig per guide Appendix B: no on-device sinf/cosf).
+        self._rope_cache: dict = {}
+
+    def _cos_sin(self, S: int, device, dtype):
+        key = (S, device, dtype)
+        hit = self._rope_cache.get(key)
+        if hit is None:
+            cos, sin = ops.torch_ref.rope_cos_sin(S, self.cfg.head_dim,
+                                                  self.cfg.rope_theta, device)
+            hit = (cos.to(dtype), sin.to(dtype))
+            self._rope_cache = {key: hit}   # keep one entry (static shapes)
+        return hit
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        x = self.embed_tokens(input_ids)
+        cos, sin = self._cos_sin(input_ids.shape[1], x.device, x.dtype)
+        for layer in self.layers:
+            x = layer(x, cos, sin)
+        return self.norm(x)
+
+
+class LlamaForCausalLM(nn.Module):
+    config_class = LlamaConfig
+
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.model = LlamaModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.cfg.initializer_range
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(0.0, std)
+            if module.bias is not None:
+                module.bias.data.zero_()
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(0.0, std)
+
+    def forward(self, input_ids: torch.Tensor,
+                labels: Optional[torch.Tensor] = None,
+                attention_mask: Optional[torch.Tensor] = None):
+        """HF-compatible call: returns (loss, logits) when labels given,
+        (logits,) otherwise — the trainer indexes outputs[0]
+        (reference trainer_decoupled.py:29-34)."""
+        hidden = self.model(input_ids)
+        logits = self.lm_head(hidden)
+        if labels is None:
+            return (logits,)
+        loss = ops.causal_lm_loss(logits, labels)
+        return (loss, logits)

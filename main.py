@@ -1,0 +1,75 @@
+"""Training entry point.
+
+CLI parity with the reference (reference main.py:25-67, README.md:52-58):
+
+    python main.py train=acco data=synthetic model=gptneo \
+        train.nb_steps_tot=100 run_name=myrun
+
+Config groups and flag surface mirror the reference's Hydra setup; the data
+default is `synthetic` because this environment has no network (use
+data=openwebtext / data=alpaca on a machine with HF hub access).
+"""
+
+import logging
+import sys
+
+import torch
+
+from acco_amd.config import load_config
+from acco_amd.data.synthetic import SyntheticCausalLMDataset
+from acco_amd.engine.trainer import DecoupledTrainer
+from acco_amd.models import build_model
+
+logging.basicConfig(stream=sys.stdout, level=logging.INFO)
+logger = logging.getLogger("acco_amd")
+
+
+def build_datasets(cfg, model):
+    data = cfg.data
+    if data.get("kind") == "synthetic":
+        vocab = model.cfg.vocab_size
+        seq = cfg.train.max_length
+        train = SyntheticCausalLMDataset(data.n_train_sequences, seq, vocab,
+                                         seed=cfg.seed)
+        evald = (SyntheticCausalLMDataset(data.n_eval_sequences, seq, vocab,
+                                          seed=cfg.seed + 1)
+                 if data.n_eval_sequences else None)
+        return train, evald, None
+    # HF path (needs network or local cache): same split policy as the
+    # reference (train_test_split 0.05 seed 42, reference main.py:50)
+    import datasets
+    from transformers import AutoTokenizer
+    ds = datasets.load_dataset(data.path)
+    ds = ds["train"].train_test_split(0.05, seed=42)
+    tokenizer = AutoTokenizer.from_pretrained(cfg.model.tokenizer)
+    tokenizer.pad_token_id = tokenizer.eos_token_id
+    return ds["train"], ds["test"], tokenizer
+
+
+def main(argv=None):
+    cfg = load_config(argv if argv is not None else sys.argv[1:])
+    torch.manual_seed(42)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(42)  # reference main.py:28
+
+    model = build_model(cfg.model)
+    logger.info("model instantiated (%.1fM params)",
+                sum(p.numel() for p in model.parameters()) / 1e6)
+    train_ds, eval_ds, tokenizer = build_datasets(cfg, model)
+
+    trainer = DecoupledTrainer(
+        model=model,
+        tokenizer=tokenizer,
+        train_dataset=train_ds,
+        eval_dataset=eval_ds,
+        text_column_name="text",
+        args=cfg.train,
+        log=logger,
+        preprocess_dataset_fn=None,
+        run_name=cfg.run_name,
+    )
+    trainer.train()
+
+
+if __name__ == "__main__":
+    main()

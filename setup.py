@@ -1,0 +1,36 @@
+"""In-tree build of the gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+produces acco_amd/_hip_ops*.so next to the package sources so the binary
+travels with the repo snapshot to GPU boxes.
+"""
+
+import glob
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+sources = sorted(glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.cpp"))
+                 + glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.hip")))
+
+setup(
+    name="acco_amd_hip_ops",
+    ext_modules=[
+        CUDAExtension(
+            name="acco_amd._hip_ops",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)

@@ -1,0 +1,290 @@
+"""The ACCO two-round state machine vs an exact sequential oracle.
+
+world_size=2 on gloo/CPU. The threaded engine records its per-round local
+grad counts (timing-dependent: slow coms let compute accumulate extra
+micro-batches — that heterogeneity is an ACCO feature, reference
+trainer_decoupled.py:85-98); the oracle replays the recorded schedule
+through a from-first-principles implementation of the two-round algebra
+(tentative step on even rounds, true step + scheduler advance on odd, grads
+zeroed after even rounds) and must reproduce the final parameters exactly
+(fp32 end to end).
+"""
+
+import os
+
+import torch
+import torch.nn as nn
+
+from tests.conftest import run_distributed
+from tests.dist_utils import init_worker, teardown_worker
+
+D = 6          # param count of the toy model
+N_ACC = 1
+TARGET = 24    # nb_grad_tot
+
+
+def make_batches(rank: int, n: int = 200):
+    g = torch.Generator().manual_seed(1000 + rank)
+    xs = torch.randn(n, 4, D, generator=g)
+    ys = torch.randn(n, 4, 1, generator=g)
+    return [(xs[i], ys[i]) for i in range(n)]
+
+
+def model_grad(w: torch.Tensor, batch) -> torch.Tensor:
+    """d/dw of mean((x @ w - y)^2), w: [D] — closed form, fp64-exactness not
+    needed since both sides use identical torch ops."""
+    x, y = batch
+    wv = w.detach().clone().requires_grad_(True)
+    loss = ((x @ wv.view(D, 1) - y) ** 2).mean()
+    loss.backward()
+    return wv.grad.detach().clone()
+
+
+def _worker_acco(rank, world, port, tmpdir, n_warmup):
+    init_worker(rank, world, port)
+    from acco_amd.engine.acco import AccoEngine
+    from acco_amd.engine.scheduler import LRSchedule
+    from acco_amd.engine.sharded_adamw import ShardedAdamW
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+
+    torch.manual_seed(7)          # same init on both ranks
+    model = nn.Linear(D, 1, bias=False)
+    device = torch.device("cpu")
+
+    from acco_amd.engine import arena
+    spec = ShardSpec.build(D, world, buckets=2, align=2)
+    params = arena.flatten_params(model, torch.float32, device,
+                                  pad_to=spec.total)
+    grads = arena.attach_grad_arena(model, torch.float32, device,
+                                    pad_to=spec.total)
+    comm = CommBackend(device)
+    opt = ShardedAdamW(spec, rank, device, lr=1e-2, betas=(0.9, 0.95),
+                       eps=1e-8, weight_decay=0.01)
+    sched = LRSchedule(1e-2, 4, TARGET, "cosine")
+
+    batches = make_batches(rank)
+    it = iter(batches)
+
+    def next_batch():
+        return next(it)
+
+    def forward_backward(batch):
+        x, y = batch
+        loss = ((x @ model.weight.t() - y) ** 2).mean()
+        loss.backward()
+        return loss.detach()
+
+    eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
+                     spec=spec, comm=comm, rank=rank, device=device, opt=opt,
+                     sched=sched, forward_backward=forward_backward,
+                     next_batch=next_batch, n_grad_accumulation=N_ACC)
+    opt.init_master_from_buffer(params)
+    eng.trace = []
+    eng.train_acco(TARGET, n_warmup_steps=n_warmup)
+
+    torch.save({
+        "params": params[:D].clone(),
+        "trace": eng.trace,
+        "round_idx": eng.round_idx,
+        "count_grad_tot": eng.count_grad_tot,
+        "w0": torch.manual_seed(7) and nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone(),
+        "sched_step": sched.current_step,
+    }, os.path.join(tmpdir, f"res_{rank}.pt"))
+    teardown_worker()
+
+
+def _adamw_math(p, m, v, step, g, lr, b1=0.9, b2=0.95, eps=1e-8, wd=0.01):
+    """torch.optim.AdamW-formulation step (pure function)."""
+    t = step + 1
+    p = p * (1 - lr * wd)
+    m = b1 * m + (1 - b1) * g
+    v = b2 * v + (1 - b2) * g * g
+    denom = v.sqrt() / (1 - b2 ** t) ** 0.5 + eps
+    p = p - lr / (1 - b1 ** t) * m / denom
+    return p, m, v
+
+
+def oracle_replay(w0, traces, world, n_warmup, lr=1e-2):
+    """Replay the ACCO algebra from first principles on the recorded
+    per-round local-count schedule."""
+    from acco_amd.engine.scheduler import LRSchedule
+    sched = LRSchedule(lr, 4, TARGET, "cosine")
+    batches = {r: make_batches(r) for r in range(world)}
+    ptr = {r: 0 for r in range(world)}
+
+    def take(r):
+        b = batches[r][ptr[r]]
+        ptr[r] += 1
+        return b
+
+    P = w0.clone()                       # broadcast params (bf16 arena ~ fp32 here)
+    master = P.clone()                   # fp32 master (full vector: oracle is unsharded)
+    m = torch.zeros(D)
+    v = torch.zeros(D)
+    step = 0
+    count_tot = 0
+
+    assert n_warmup == 0, "oracle covers the n_warmup=0 path"
+    # bootstrap (reference prepare_grads/prepare_buffer_com, engine.bootstrap)
+    accum = {}
+    buffer = {}
+    count_local = {}
+    count_round = {}
+    for r in range(world):
+        g = model_grad(P, take(r))
+        accum[r] = g.clone()
+        buffer[r] = g.clone()
+        count_local[r] = 1
+        count_round[r] = 1
+
+    n_rounds = len(traces[0])
+    for idx in range(n_rounds):
+        commit = idx % 2 == 1
+        # --- com round idx (uses buffer/count_round set before the round)
+        G = sum(buffer[r] for r in range(world))
+        cnt = sum(count_round[r] for r in range(world))
+        cur_lr = sched.lr()
+        newP, newM, newV = _adamw_math(master, m, v, step, G / cnt, cur_lr)
+        if commit:
+            master, m, v = newP, newM, newV
+            step += 1
+            count_tot += cnt
+            sched.advance(cnt)
+        NP = newP.clone()
+        # --- compute during round idx: replay recorded new micro-batches
+        for r in range(world):
+            k_new = traces[r][idx] - count_local[r]
+            assert k_new >= 0
+            for _ in range(k_new):
+                accum[r] = accum[r] + model_grad(P, take(r))
+            count_local[r] = traces[r][idx]
+        # --- buffer update after round idx
+        P = NP
+        for r in range(world):
+            buffer[r] = accum[r].clone()
+            count_round[r] = count_local[r]
+            if idx % 2 == 0:
+                accum[r] = torch.zeros(D)
+                count_local[r] = 0
+    return P, count_tot
+
+
+def test_acco_matches_oracle_ws2():
+    tmpdir = run_distributed(_worker_acco, 2, args=(0,), timeout=240)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    # both ranks end with identical params
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert res[0]["round_idx"] == res[1]["round_idx"]
+    assert res[0]["count_grad_tot"] >= TARGET
+
+    torch.manual_seed(7)
+    w0 = nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone()
+    traces = {r: res[r]["trace"] for r in range(2)}
+    assert len(traces[0]) == len(traces[1]) == res[0]["round_idx"]
+
+    P, count_tot = oracle_replay(w0, traces, 2, 0)
+    assert count_tot == res[0]["count_grad_tot"]
+    assert torch.allclose(P, res[0]["params"], atol=1e-6, rtol=1e-6), \
+        (P, res[0]["params"])
+
+
+def _worker_dpu(rank, world, port, tmpdir):
+    init_worker(rank, world, port)
+    _run_mode(rank, world, tmpdir, mode="dpu")
+    teardown_worker()
+
+
+def _run_mode(rank, world, tmpdir, mode):
+    from acco_amd.engine.acco import AccoEngine
+    from acco_amd.engine.scheduler import LRSchedule
+    from acco_amd.engine.sharded_adamw import ShardedAdamW
+    from acco_amd.engine import arena
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+
+    torch.manual_seed(7)
+    model = nn.Linear(D, 1, bias=False)
+    device = torch.device("cpu")
+    spec = ShardSpec.build(D, world, buckets=2, align=2)
+    params = arena.flatten_params(model, torch.float32, device, pad_to=spec.total)
+    grads = arena.attach_grad_arena(model, torch.float32, device, pad_to=spec.total)
+    comm = CommBackend(device)
+    opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
+    sched = LRSchedule(1e-2, 4, TARGET, "cosine")
+    batches = make_batches(rank)
+    it = iter(batches)
+
+    def forward_backward(batch):
+        x, y = batch
+        loss = ((x @ model.weight.t() - y) ** 2).mean()
+        loss.backward()
+        return loss.detach()
+
+    eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
+                     spec=spec, comm=comm, rank=rank, device=device, opt=opt,
+                     sched=sched, forward_backward=forward_backward,
+                     next_batch=lambda: next(it), n_grad_accumulation=N_ACC)
+    opt.init_master_from_buffer(params)
+    if mode == "dpu":
+        eng.train_dpu(TARGET, n_warmup_steps=0)
+    torch.save({"params": params[:D].clone(),
+                "count": eng.count_grad_tot},
+               os.path.join(tmpdir, f"res_{rank}.pt"))
+
+
+def test_dpu_runs_and_syncs_ws2():
+    tmpdir = run_distributed(_worker_dpu, 2, timeout=240)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert res[0]["count"] >= TARGET
+    assert torch.isfinite(res[0]["params"]).all()
+
+
+def _worker_acco_warmup(rank, world, port, tmpdir):
+    init_worker(rank, world, port)
+    _run_warmup(rank, world, tmpdir)
+    teardown_worker()
+
+
+def _run_warmup(rank, world, tmpdir):
+    from acco_amd.engine.acco import AccoEngine
+    from acco_amd.engine.scheduler import LRSchedule
+    from acco_amd.engine.sharded_adamw import ShardedAdamW
+    from acco_amd.engine import arena
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+
+    torch.manual_seed(7)
+    model = nn.Linear(D, 1, bias=False)
+    device = torch.device("cpu")
+    spec = ShardSpec.build(D, world, buckets=2, align=2)
+    params = arena.flatten_params(model, torch.float32, device, pad_to=spec.total)
+    grads = arena.attach_grad_arena(model, torch.float32, device, pad_to=spec.total)
+    comm = CommBackend(device)
+    opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
+    sched = LRSchedule(1e-2, 4, TARGET, "cosine")
+    it = iter(make_batches(rank))
+
+    def forward_backward(batch):
+        x, y = batch
+        loss = ((x @ model.weight.t() - y) ** 2).mean()
+        loss.backward()
+        return loss.detach()
+
+    eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
+                     spec=spec, comm=comm, rank=rank, device=device, opt=opt,
+                     sched=sched, forward_backward=forward_backward,
+                     next_batch=lambda: next(it), n_grad_accumulation=N_ACC)
+    opt.init_master_from_buffer(params)
+    eng.train_acco(TARGET, n_warmup_steps=3)
+    torch.save({"params": params[:D].clone(), "count": eng.count_grad_tot},
+               os.path.join(tmpdir, f"res_{rank}.pt"))
+
+
+def test_acco_warmup_path_ws2():
+    tmpdir = run_distributed(_worker_acco_warmup, 2, timeout=240)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert torch.isfinite(res[0]["params"]).all()
+    assert res[0]["count"] >= TARGET

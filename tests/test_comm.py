@@ -1,0 +1,82 @@
+"""ShardSpec geometry + gloo collectives semantics (world_size=2 on CPU)."""
+
+import os
+
+import torch
+
+from acco_amd.parallel.comm import ShardSpec
+
+from tests.conftest import run_distributed
+from tests.dist_utils import init_worker, teardown_worker
+
+
+def test_shardspec_geometry():
+    spec = ShardSpec.build(n=10_000, world=8, buckets=8, align=256)
+    assert spec.total >= 10_000
+    assert spec.seg % 256 == 0
+    assert spec.total == spec.nb * spec.world * spec.seg
+    assert spec.owned == spec.nb * spec.seg
+    buf = torch.arange(spec.total, dtype=torch.float32)
+    seen = torch.zeros(spec.total, dtype=torch.bool)
+    for j in range(spec.nb):
+        b = spec.bucket_view(buf, j)
+        assert b.numel() == spec.bucket_elems
+        for r in range(spec.world):
+            s = spec.seg_view(buf, j, r)
+            assert s.numel() == spec.seg
+            lo = int(s[0].item())
+            seen[lo:lo + spec.seg] = True
+    assert seen.all()
+
+
+def test_shardspec_small_n():
+    spec = ShardSpec.build(n=10, world=2, buckets=4, align=4)
+    assert spec.seg == 4 and spec.total == 32
+
+
+def _worker_collectives(rank, world, port, tmpdir):
+    init_worker(rank, world, port)
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+    spec = ShardSpec.build(n=60, world=world, buckets=3, align=4)
+    comm = CommBackend(torch.device("cpu"))
+
+    # buffer filled with rank-dependent values
+    buf = torch.full((spec.total,), float(rank + 1))
+    works = [comm.reduce_scatter_bucket_async(buf, spec, j, rank)
+             for j in range(spec.nb)]
+    for w in works:
+        w.wait()
+    expected_sum = sum(r + 1 for r in range(world))
+    for j in range(spec.nb):
+        seg = spec.seg_view(buf, j, rank)
+        assert torch.all(seg == expected_sum), (j, seg)
+
+    # each rank writes its id into its segments, all-gather broadcasts
+    for j in range(spec.nb):
+        spec.seg_view(buf, j, rank).fill_(float(100 + rank))
+    ag = [comm.all_gather_bucket_async(buf, spec, j, rank)
+          for j in range(spec.nb)]
+    for w in ag:
+        w.wait()
+    for j in range(spec.nb):
+        for r in range(world):
+            assert torch.all(spec.seg_view(buf, j, r) == float(100 + r))
+
+    # async count all-reduce
+    c = torch.tensor([rank + 1], dtype=torch.int32)
+    comm.all_reduce_sum_async(c).wait()
+    assert int(c) == expected_sum
+
+    # avg all-reduce
+    t = torch.full((8,), float(rank))
+    comm.all_reduce_avg(t)
+    assert torch.allclose(t, torch.full((8,), (world - 1) / 2))
+
+    torch.save(torch.tensor(1), os.path.join(tmpdir, f"ok_{rank}.pt"))
+    teardown_worker()
+
+
+def test_gloo_bucket_collectives_ws2():
+    tmpdir = run_distributed(_worker_collectives, 2)
+    for r in range(2):
+        assert os.path.exists(os.path.join(tmpdir, f"ok_{r}.pt"))

@@ -1,0 +1,86 @@
+"""NativeZeroDDP (bucketed reduce-scatter + ZeRO-1 fused AdamW + all-gather)
+vs a plain single-process AdamW oracle on averaged gradients."""
+
+import os
+
+import torch
+import torch.nn as nn
+
+from tests.conftest import run_distributed
+from tests.dist_utils import init_worker, teardown_worker
+
+D_IN, D_OUT = 6, 3
+STEPS = 5
+N_ACC = 2
+LR = 1e-2
+
+
+def make_model():
+    torch.manual_seed(11)
+    return nn.Linear(D_IN, D_OUT, bias=True)
+
+
+def make_batch(rank: int, step: int, micro: int):
+    g = torch.Generator().manual_seed(rank * 10007 + step * 101 + micro)
+    x = torch.randn(5, D_IN, generator=g)
+    y = torch.randn(5, D_OUT, generator=g)
+    return x, y
+
+
+def loss_fn(model, batch):
+    x, y = batch
+    return ((model(x) - y) ** 2).mean()
+
+
+def _worker_ddp(rank, world, port, tmpdir):
+    init_worker(rank, world, port)
+    from acco_amd.engine import arena
+    from acco_amd.engine.sharded_adamw import ShardedAdamW
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+    from acco_amd.parallel.ddp import NativeZeroDDP
+
+    model = make_model()
+    device = torch.device("cpu")
+    n = arena.live_numel(model)
+    spec = ShardSpec.build(n, world, buckets=2, align=2)
+    params = arena.flatten_params(model, torch.float32, device, pad_to=spec.total)
+    grads = arena.attach_grad_arena(model, torch.float32, device, pad_to=spec.total)
+    comm = CommBackend(device)
+    opt = ShardedAdamW(spec, rank, device, lr=LR, weight_decay=0.01)
+    opt.init_master_from_buffer(params)
+    ddp = NativeZeroDDP(model, params, grads, n, spec, comm, rank, opt)
+
+    for step in range(STEPS):
+        for micro in range(N_ACC):
+            if micro == N_ACC - 1:
+                ddp.begin_sync_microbatch()
+            loss = loss_fn(model, make_batch(rank, step, micro)) / N_ACC
+            loss.backward()
+        ddp.finish_step(grad_scale=1.0 / world)
+        ddp.zero_grad()
+
+    torch.save(params[:n].clone(), os.path.join(tmpdir, f"p_{rank}.pt"))
+    teardown_worker()
+
+
+def test_native_ddp_matches_adamw_oracle_ws2():
+    world = 2
+    tmpdir = run_distributed(_worker_ddp, world, timeout=180)
+    res = [torch.load(os.path.join(tmpdir, f"p_{r}.pt"), weights_only=False)
+           for r in range(world)]
+    assert torch.equal(res[0], res[1])
+
+    # oracle: single-process AdamW on rank-averaged accumulated grads
+    model = make_model()
+    opt = torch.optim.AdamW(model.parameters(), lr=LR, betas=(0.9, 0.95),
+                            eps=1e-8, weight_decay=0.01)
+    for step in range(STEPS):
+        opt.zero_grad()
+        for rank in range(world):
+            for micro in range(N_ACC):
+                loss = loss_fn(model, make_batch(rank, step, micro))
+                (loss / (N_ACC * world)).backward()
+        opt.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    assert torch.allclose(res[0], flat, atol=1e-6, rtol=1e-5), \
+        (res[0] - flat).abs().max()

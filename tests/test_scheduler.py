@@ -1,0 +1,46 @@
+import math
+
+import pytest
+
+from acco_amd.engine.scheduler import LRSchedule
+
+
+def test_warmup_ramp():
+    s = LRSchedule(1.0, 10, 100, "cosine")
+    assert s.factor(0) == 0.0
+    assert s.factor(5) == pytest.approx(0.5)
+    assert s.factor(10) == pytest.approx(1.0)
+
+
+def test_cosine_decay():
+    s = LRSchedule(2.0, 0, 100, "cosine")
+    assert s.factor(0) == pytest.approx(1.0)
+    assert s.factor(50) == pytest.approx(0.5)
+    assert s.factor(100) == pytest.approx(0.0, abs=1e-12)
+    s.current_step = 50
+    assert s.lr() == pytest.approx(1.0)
+
+
+def test_advance_in_batches_matches_hf_semantics():
+    # the reference advances by the global grad count per com round
+    # (scheduler.step() + _step_count += count-1, trainer_decoupled.py:102-104)
+    a = LRSchedule(1.0, 5, 50, "cosine")
+    b = LRSchedule(1.0, 5, 50, "cosine")
+    a.advance(7)
+    for _ in range(7):
+        b.advance(1)
+    assert a.lr() == pytest.approx(b.lr())
+
+
+def test_linear_and_constant():
+    lin = LRSchedule(1.0, 0, 10, "linear")
+    assert lin.factor(5) == pytest.approx(0.5)
+    const = LRSchedule(1.0, 2, 10, "constant")
+    assert const.factor(7) == 1.0
+    assert const.factor(1) == pytest.approx(0.5)
+
+
+def test_clamps_past_total():
+    s = LRSchedule(1.0, 0, 10, "cosine")
+    assert s.factor(20) == pytest.approx(0.0, abs=1e-12)
+    assert not math.isnan(s.factor(20))

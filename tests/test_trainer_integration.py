@@ -1,0 +1,84 @@
+"""End-to-end DecoupledTrainer runs on CPU/gloo world_size=2 (BASELINE.json
+config 1: GPT-2-small-shaped train=ddp on synthetic data; plus acco)."""
+
+import os
+
+import torch
+
+from tests.conftest import run_distributed
+from tests.dist_utils import teardown_worker
+
+
+def _run_trainer(rank, world, port, tmpdir, method):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.chdir(tmpdir)
+
+    from acco_amd.config import load_config
+    from acco_amd.data.synthetic import SyntheticCausalLMDataset
+    from acco_amd.engine.trainer import DecoupledTrainer
+    from acco_amd.models import GPTNeoConfig, GPTNeoForCausalLM
+
+    cfg = load_config([
+        f"train={method}", "data=synthetic", "model=gptneo",
+        "train.nb_steps_tot=8", "train.batch_size=2", "train.max_length=16",
+        "train.use_mixed_precision=false", "train.save=false",
+        "train.warmup=2", "train.n_warmup_steps=0",
+        "train.dataloader_num_workers=0", "train.comm_buckets=2",
+        "train.dataloader_persistent_workers=false",
+    ])
+    torch.manual_seed(42)
+    mcfg = GPTNeoConfig(hidden_size=32, num_layers=2, num_heads=2,
+                        vocab_size=64, max_position_embeddings=32,
+                        window_size=8)
+    model = GPTNeoForCausalLM(mcfg)
+    train_ds = SyntheticCausalLMDataset(32, 16, 64, seed=5 + rank)
+
+    trainer = DecoupledTrainer(model=model, train_dataset=train_ds,
+                               eval_dataset=None, args=cfg.train,
+                               run_name="itest")
+    trainer.train()
+
+    flat = trainer.params[:trainer.n_live].clone()
+    torch.save({"params": flat}, os.path.join(tmpdir, f"p_{method}_{rank}.pt"))
+    teardown_worker()
+
+
+def _worker_ddp(rank, world, port, tmpdir):
+    _run_trainer(rank, world, port, tmpdir, "ddp")
+
+
+def _worker_acco(rank, world, port, tmpdir):
+    _run_trainer(rank, world, port, tmpdir, "acco")
+
+
+def _worker_dpu(rank, world, port, tmpdir):
+    _run_trainer(rank, world, port, tmpdir, "dpu")
+
+
+def _check(tmpdir, method, world=2):
+    res = [torch.load(os.path.join(tmpdir, f"p_{method}_{r}.pt"),
+                      weights_only=False) for r in range(world)]
+    assert torch.equal(res[0]["params"], res[1]["params"]), \
+        f"{method}: ranks diverged"
+    assert torch.isfinite(res[0]["params"]).all()
+    # results.csv written by rank 0
+    assert os.path.exists(os.path.join(tmpdir, "results.csv"))
+
+
+def test_trainer_ddp_ws2():
+    tmpdir = run_distributed(_worker_ddp, 2, timeout=300)
+    _check(tmpdir, "ddp")
+
+
+def test_trainer_acco_ws2():
+    tmpdir = run_distributed(_worker_acco, 2, timeout=300)
+    _check(tmpdir, "acco")
+
+
+def test_trainer_dpu_ws2():
+    tmpdir = run_distributed(_worker_dpu, 2, timeout=300)
+    _check(tmpdir, "dpu")
