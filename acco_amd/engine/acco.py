@@ -97,6 +97,11 @@ class AccoEngine:
         # test instrumentation: when set to a list, records the per-round
         # LOCAL grad count at each buffer update (oracle replay support)
         self.trace: Optional[list] = None
+        # bench instrumentation: called on EVERY rank after each com-round
+        # boundary (round_idx already incremented); python-side micro-batch
+        # counter for token accounting
+        self.on_round_boundary: Optional[Callable[[int], None]] = None
+        self.micro_steps = 0
 
     # ------------------------------------------------------------ pieces
 
@@ -105,6 +110,7 @@ class AccoEngine:
         (reference gradient_step :18-39)."""
         loss = self.forward_backward(inputs)
         self.count_grad_local += 1
+        self.micro_steps += 1
         self.loss_static.copy_(loss.detach().float().reshape(1))
 
     @torch.no_grad()
@@ -191,16 +197,22 @@ class AccoEngine:
 
     # --------------------------------------------------------- ACCO loop
 
-    def train_acco(self, nb_grad_tot: int, n_warmup_steps: int = 0) -> None:
+    def train_acco(self, nb_grad_tot: int, n_warmup_steps: int = 0,
+                   max_rounds: Optional[int] = None) -> None:
         self.bootstrap(n_warmup_steps)
         barrier = threading.Barrier(2)
         stop = threading.Event()
+
+        def keep_going() -> bool:
+            if max_rounds is not None and self.round_idx >= max_rounds:
+                return False
+            return self.count_grad_tot < nb_grad_tot
 
         def com_routine():
             try:
                 self.com_stream.wait_default(self.device)
                 with self.com_stream.activate():
-                    while self.count_grad_tot < nb_grad_tot and not stop.is_set():
+                    while keep_going() and not stop.is_set():
                         commit = (self.round_idx % 2 == 1)
                         self.communication_round(commit=commit)
                         self.com_event.record_and_sync(self.com_stream)
@@ -219,7 +231,7 @@ class AccoEngine:
         try:
             self.grad_stream.wait_default(self.device)
             with self.grad_stream.activate():
-                while self.count_grad_tot < nb_grad_tot:
+                while keep_going():
                     for _ in range(self.n_acc):
                         self.gradient_step(self.next_batch())
                     self.end_of_grad.record_and_sync(self.grad_stream)
@@ -233,6 +245,8 @@ class AccoEngine:
                         self.update_event.record_and_sync(self.grad_stream)
                         self.round_idx += 1
                         barrier.wait()
+                        if self.on_round_boundary is not None:
+                            self.on_round_boundary(self.round_idx)
                         if self.on_round_complete is not None and self.rank == 0:
                             self.on_round_complete(self.round_idx,
                                                    self.count_grad_tot)

@@ -48,18 +48,31 @@ def ext_available() -> bool:
     return _load_ext() is not None
 
 
-def _use_ref(t: torch.Tensor) -> bool:
+def have_kernel(name: str) -> bool:
+    ext = _load_ext()
+    return ext is not None and hasattr(ext, name)
+
+
+def _use_ref(t: torch.Tensor, kernel: str = "") -> bool:
+    """torch-reference path when: CPU tensor, forced via env, or the HIP
+    kernel for this op has not been built yet. On a GPU box with the
+    extension entirely missing, ops fail loudly instead (hip_ext raises) —
+    the kernel registry only tolerates per-op gaps during bring-up."""
     if not t.is_cuda:
         return True
     if os.environ.get("ACCO_FORCE_REF") == "1":
         return True
+    if kernel:
+        if not ext_available():
+            hip_ext()  # raises with the build instructions
+        return not have_kernel(kernel)
     return False
 
 
 # ---------------------------------------------------------------- model ops
 
 def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
-    if _use_ref(x):
+    if _use_ref(x, "rmsnorm_fwd"):
         return torch_ref.rms_norm(x, weight, eps)
     from acco_amd.ops.autograd import RMSNormFn
     return RMSNormFn.apply(x, weight, eps)
@@ -67,21 +80,21 @@ def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
 
 def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
                eps: float) -> torch.Tensor:
-    if _use_ref(x):
+    if _use_ref(x, "layernorm_fwd"):
         return torch_ref.layer_norm(x, weight, bias, eps)
     from acco_amd.ops.autograd import LayerNormFn
     return LayerNormFn.apply(x, weight, bias, eps)
 
 
 def gelu_new(x: torch.Tensor) -> torch.Tensor:
-    if _use_ref(x):
+    if _use_ref(x, "gelu_fwd"):
         return torch_ref.gelu_new(x)
     from acco_amd.ops.autograd import GeluNewFn
     return GeluNewFn.apply(x)
 
 
 def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
-    if _use_ref(gate):
+    if _use_ref(gate, "swiglu_fwd"):
         return torch_ref.swiglu(gate, up)
     from acco_amd.ops.autograd import SwiGLUFn
     return SwiGLUFn.apply(gate, up)
@@ -89,7 +102,7 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
 
 def rope_apply(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
                sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    if _use_ref(q):
+    if _use_ref(q, "rope_fwd"):
         return torch_ref.rope_apply(q, k, cos, sin)
     from acco_amd.ops.autograd import RoPEFn
     return RoPEFn.apply(q, k, cos, sin)
@@ -98,14 +111,14 @@ def rope_apply(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
 def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      scale: Optional[float] = None,
                      window: Optional[int] = None) -> torch.Tensor:
-    if _use_ref(q):
+    if _use_ref(q, "attn_fwd"):
         return torch_ref.causal_attention(q, k, v, scale=scale, window=window)
     from acco_amd.ops.autograd import AttentionFn
     return AttentionFn.apply(q, k, v, scale, window)
 
 
 def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-    if _use_ref(logits):
+    if _use_ref(logits, "ce_fwd"):
         return torch_ref.causal_lm_loss(logits, labels)
     from acco_amd.ops.autograd import CausalLMLossFn
     return CausalLMLossFn.apply(logits, labels)
