@@ -37,13 +37,12 @@ class GPTNeoSelfAttention(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, d = x.shape
         H, hd = self.cfg.num_heads, self.cfg.head_dim
-        q = self.q_proj(x).view(B, S, H, hd).transpose(1, 2)
-        k = self.k_proj(x).view(B, S, H, hd).transpose(1, 2)
-        v = self.v_proj(x).view(B, S, H, hd).transpose(1, 2)
+        q = self.q_proj(x).view(B, S, H, hd)
+        k = self.k_proj(x).view(B, S, H, hd)
+        v = self.v_proj(x).view(B, S, H, hd)
         window = self.cfg.window_size if self.attention_type == "local" else None
         o = ops.causal_attention(q, k, v, scale=1.0, window=window)
-        o = o.transpose(1, 2).reshape(B, S, d)
-        return self.out_proj(o)
+        return self.out_proj(o.reshape(B, S, d))
 
 
 class GPTNeoAttention(nn.Module):

@@ -45,13 +45,14 @@ class LlamaAttention(nn.Module):
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
         cfg = self.cfg
-        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim).transpose(1, 2)
-        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2)
-        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2)
+        # keep everything in the projections' natural [B, S, H, D] layout:
+        # RoPE and attention kernels walk D coalesced, no transpose copies
+        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
+        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
         q, k = ops.rope_apply(q, k, cos, sin)
-        o = ops.causal_attention(q, k, v)          # [B, H, S, D]
-        o = o.transpose(1, 2).reshape(B, S, -1)
-        return self.o_proj(o)
+        o = ops.causal_attention(q, k, v)          # [B, S, H, D]
+        return self.o_proj(o.reshape(B, S, -1))
 
 
 class LlamaMLP(nn.Module):
@@ -90,19 +91,19 @@ class LlamaModel(nn.Module):
         # (host-precomputed trig per guide Appendix B: no on-device sinf/cosf).
         self._rope_cache: dict = {}
 
-    def _cos_sin(self, S: int, device, dtype):
-        key = (S, device, dtype)
+    def _cos_sin(self, S: int, device):
+        key = (S, device)
         hit = self._rope_cache.get(key)
         if hit is None:
-            cos, sin = ops.torch_ref.rope_cos_sin(S, self.cfg.head_dim,
-                                                  self.cfg.rope_theta, device)
-            hit = (cos.to(dtype), sin.to(dtype))
+            # fp32 tables: the RoPE HIP kernel multiplies in fp32
+            hit = ops.torch_ref.rope_cos_sin(S, self.cfg.head_dim,
+                                             self.cfg.rope_theta, device)
             self._rope_cache = {key: hit}   # keep one entry (static shapes)
         return hit
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         x = self.embed_tokens(input_ids)
-        cos, sin = self._cos_sin(input_ids.shape[1], x.device, x.dtype)
+        cos, sin = self._cos_sin(input_ids.shape[1], x.device)
         for layer in self.layers:
             x = layer(x, cos, sin)
         return self.norm(x)

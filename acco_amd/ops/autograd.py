@@ -1,0 +1,126 @@
+"""torch.autograd.Function wrappers around the gfx950 HIP kernels.
+
+Only imported on the GPU path (ops/__init__ dispatch); each Function pairs a
+hand-written forward kernel with its hand-written backward kernel, with
+fp32 row statistics saved between them. Numerics tests compare each against
+the fp32 torch reference (tests/test_gpu_kernels.py).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from acco_amd import ops
+
+
+class SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        gate = gate.contiguous()
+        up = up.contiguous()
+        ctx.save_for_backward(gate, up)
+        return ops.hip_ext().swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, dout):
+        gate, up = ctx.saved_tensors
+        dg, du = ops.hip_ext().swiglu_bwd(dout.contiguous(), gate, up)
+        return dg, du
+
+
+class GeluNewFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return ops.hip_ext().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (x,) = ctx.saved_tensors
+        return ops.hip_ext().gelu_bwd(dout.contiguous(), x)
+
+
+class RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        x = x.contiguous()
+        y, rstd = ops.hip_ext().rmsnorm_fwd(x, weight.contiguous(), eps)
+        ctx.save_for_backward(x, weight, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd = ctx.saved_tensors
+        dx, dw = ops.hip_ext().rmsnorm_bwd(dy.contiguous(), x,
+                                           weight.contiguous(), rstd)
+        return dx, dw.to(weight.dtype), None
+
+
+class LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        x = x.contiguous()
+        y, mean, rstd = ops.hip_ext().layernorm_fwd(x, weight.contiguous(),
+                                                    bias.contiguous(), eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ops.hip_ext().layernorm_bwd(dy.contiguous(), x,
+                                                 weight.contiguous(), mean,
+                                                 rstd)
+        return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+class RoPEFn(torch.autograd.Function):
+    """q: [B,S,H,D], k: [B,S,Hkv,D]; cos/sin fp32 [S,D] host-precomputed."""
+
+    @staticmethod
+    def forward(ctx, q, k, cos, sin):
+        cos = cos.float().contiguous()
+        sin = sin.float().contiguous()
+        ext = ops.hip_ext()
+        q2 = ext.rope_fwd(q.contiguous(), cos, sin, False)
+        k2 = ext.rope_fwd(k.contiguous(), cos, sin, False)
+        ctx.save_for_backward(cos, sin)
+        return q2, k2
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        cos, sin = ctx.saved_tensors
+        ext = ops.hip_ext()
+        dq0 = ext.rope_fwd(dq.contiguous(), cos, sin, True)
+        dk0 = ext.rope_fwd(dk.contiguous(), cos, sin, True)
+        return dq0, dk0, None, None
+
+
+class CausalLMLossFn(torch.autograd.Function):
+    """Fused shifted CE: logits [B,S,V] bf16, labels [B,S] int64."""
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        logits = logits.contiguous()
+        labels = labels.contiguous()
+        acc, lse = ops.hip_ext().ce_fwd(logits, labels)
+        ctx.save_for_backward(logits, labels, lse, acc)
+        return acc[0] / acc[1].clamp(min=1.0)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        logits, labels, lse, acc = ctx.saved_tensors
+        dlogits = ops.hip_ext().ce_bwd(logits, labels, lse, acc,
+                                       float(grad_out))
+        return dlogits, None
+
+
+class AttentionFn(torch.autograd.Function):
+    """Flash-style causal attention (gfx950 MFMA) — kernel lands next;
+    dispatch guards on have_kernel('attn_fwd') so this is unreachable
+    until then."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale, window):
+        raise NotImplementedError("attn_fwd kernel not built yet")

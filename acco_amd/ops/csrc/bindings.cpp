@@ -10,6 +10,31 @@ extern "C" void acco_fused_adamw_launch(
     long long n, bool buf_is_bf16, bool commit,
     float scale, float lr, float beta1, float beta2, float eps,
     float weight_decay, long long step_plus_1, hipStream_t stream);
+extern "C" void acco_swiglu_fwd(const void*, const void*, void*, long long,
+                                hipStream_t);
+extern "C" void acco_swiglu_bwd(const void*, const void*, const void*, void*,
+                                void*, long long, hipStream_t);
+extern "C" void acco_gelu_fwd(const void*, void*, long long, hipStream_t);
+extern "C" void acco_gelu_bwd(const void*, const void*, void*, long long,
+                              hipStream_t);
+extern "C" void acco_rmsnorm_fwd(const void*, const void*, void*, void*,
+                                 long long, int, float, hipStream_t);
+extern "C" void acco_rmsnorm_bwd(const void*, const void*, const void*,
+                                 const void*, void*, void*, long long, int,
+                                 hipStream_t);
+extern "C" void acco_layernorm_fwd(const void*, const void*, const void*,
+                                   void*, void*, void*, long long, int, float,
+                                   hipStream_t);
+extern "C" void acco_layernorm_bwd(const void*, const void*, const void*,
+                                   const void*, const void*, void*, void*,
+                                   void*, long long, int, hipStream_t);
+extern "C" void acco_rope(const void*, void*, const float*, const float*,
+                          long long, int, int, int, bool, hipStream_t);
+extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
+                            long long, int, int, hipStream_t);
+extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
+                            void*, const float*, float, long long, int, int,
+                            hipStream_t);
 
 namespace {
 
@@ -43,11 +68,169 @@ void fused_adamw(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                           step + 1, stream.stream());
 }
 
+#define CHECK_BF16_CONTIG(t) \
+  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == at::kBFloat16 && \
+              (t).is_contiguous(), #t " must be contiguous CUDA bf16")
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---- SwiGLU / gelu_new
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u) {
+  CHECK_BF16_CONTIG(g); CHECK_BF16_CONTIG(u);
+  TORCH_CHECK(g.numel() == u.numel() && g.numel() % 8 == 0);
+  auto out = at::empty_like(g);
+  acco_swiglu_fwd(g.data_ptr(), u.data_ptr(), out.data_ptr(), g.numel(),
+                  cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor g, at::Tensor u) {
+  CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(g); CHECK_BF16_CONTIG(u);
+  auto dg = at::empty_like(g);
+  auto du = at::empty_like(u);
+  acco_swiglu_bwd(dout.data_ptr(), g.data_ptr(), u.data_ptr(), dg.data_ptr(),
+                  du.data_ptr(), g.numel(), cur_stream());
+  return {dg, du};
+}
+
+at::Tensor gelu_fwd(at::Tensor x) {
+  CHECK_BF16_CONTIG(x);
+  TORCH_CHECK(x.numel() % 8 == 0);
+  auto out = at::empty_like(x);
+  acco_gelu_fwd(x.data_ptr(), out.data_ptr(), x.numel(), cur_stream());
+  return out;
+}
+
+at::Tensor gelu_bwd(at::Tensor dout, at::Tensor x) {
+  CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(x);
+  auto dx = at::empty_like(x);
+  acco_gelu_bwd(dout.data_ptr(), x.data_ptr(), dx.data_ptr(), x.numel(),
+                cur_stream());
+  return dx;
+}
+
+// ---- RMSNorm
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w);
+  const int D = (int)w.numel();
+  TORCH_CHECK(D % 8 == 0 && D <= 16384 && x.numel() % D == 0);
+  const long long R = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  acco_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), rstd.data_ptr(),
+                   R, D, (float)eps, cur_stream());
+  return {y, rstd};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor rstd) {
+  CHECK_BF16_CONTIG(dy); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w);
+  const int D = (int)w.numel();
+  const long long R = x.numel() / D;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
+  acco_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                   rstd.data_ptr(), dx.data_ptr(), dw.data_ptr(), R, D,
+                   cur_stream());
+  return {dx, dw};
+}
+
+// ---- LayerNorm
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
+                                      double eps) {
+  CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w); CHECK_BF16_CONTIG(b);
+  const int D = (int)w.numel();
+  TORCH_CHECK(D % 8 == 0 && D <= 16384 && x.numel() % D == 0);
+  const long long R = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({R}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  acco_layernorm_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+                     mean.data_ptr(), rstd.data_ptr(), R, D, (float)eps,
+                     cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
+                                      at::Tensor w, at::Tensor mean,
+                                      at::Tensor rstd) {
+  CHECK_BF16_CONTIG(dy); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w);
+  const int D = (int)w.numel();
+  const long long R = x.numel() / D;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({D}, x.options().dtype(at::kFloat));
+  acco_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
+                     dw.data_ptr(), db.data_ptr(), R, D, cur_stream());
+  return {dx, dw, db};
+}
+
+// ---- RoPE ([B, S, H, D] contiguous)
+at::Tensor rope_fwd(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
+                    bool bwd) {
+  CHECK_BF16_CONTIG(x);
+  TORCH_CHECK(x.dim() == 4);
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat && cos_t.is_contiguous());
+  TORCH_CHECK(sin_t.scalar_type() == at::kFloat && sin_t.is_contiguous());
+  const long long B = x.size(0);
+  const int S = (int)x.size(1), H = (int)x.size(2), D = (int)x.size(3);
+  TORCH_CHECK(D % 8 == 0 && cos_t.size(0) >= S && cos_t.size(1) == D);
+  auto y = at::empty_like(x);
+  acco_rope(x.data_ptr(), y.data_ptr(), cos_t.data_ptr<float>(),
+            sin_t.data_ptr<float>(), B, S, H, D, bwd, cur_stream());
+  return y;
+}
+
+// ---- fused shifted causal-LM CE
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(logits.dim() == 3 && labels.dim() == 2);
+  TORCH_CHECK(labels.scalar_type() == at::kLong && labels.is_contiguous());
+  const long long Bn = logits.size(0);
+  const int S = (int)logits.size(1), V = (int)logits.size(2);
+  const long long T = Bn * S;
+  auto lse = at::empty({T}, logits.options().dtype(at::kFloat));
+  auto acc = at::zeros({2}, logits.options().dtype(at::kFloat));
+  acco_ce_fwd(logits.data_ptr(),
+              reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
+              lse.data_ptr<float>(), acc.data_ptr<float>(), T, S, V,
+              cur_stream());
+  return {acc, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
+                  at::Tensor acc, double dloss) {
+  CHECK_BF16_CONTIG(logits);
+  const long long Bn = logits.size(0);
+  const int S = (int)logits.size(1), V = (int)logits.size(2);
+  const long long T = Bn * S;
+  auto dlogits = at::empty_like(logits);
+  acco_ce_bwd(logits.data_ptr(),
+              reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
+              lse.data_ptr<float>(), dlogits.data_ptr(),
+              acc.data_ptr<float>(), (float)dloss, T, S, V, cur_stream());
+  return dlogits;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw,
         "Fused sharded AdamW (gfx950): cast+scale+AdamW+bf16 writeout; "
         "commit=false = ACCO tentative step");
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
   m.attr("_gfx950") = true;
 }

@@ -1,0 +1,159 @@
+#include "hip/hip_runtime.h"
+// Fused shifted causal-LM cross-entropy for gfx950 (SURVEY.md §2.5 K1 loss
+// + K9). The eager path materializes fp32 log-softmax over [B,S,V] (vocab
+// 50k: ~1.6 GB of HBM traffic per cast); this kernel does one online
+// logsumexp pass over bf16 logits (fwd) and one fused softmax-minus-onehot
+// pass (bwd), both at HBM-roofline.
+//
+// Forward: one 256-thread block per token (b, s<S-1), label = labels[b,s+1];
+// emits per-token lse into [B,S] fp32 (bwd recompute support), and
+// atomicAdds {loss_sum, n_valid} into a 2-elem fp32 accumulator.
+// ignore_index = -100 tokens contribute nothing.
+
+#include "common.h"
+
+namespace {
+
+using u16 = unsigned short;
+constexpr int BLOCK = 256;
+
+__global__ void ce_fwd_kernel(const u16* __restrict__ logits,  // [T, V]
+                              const long long* __restrict__ labels,  // [T]
+                              float* __restrict__ lse,         // [T]
+                              float* __restrict__ acc,         // {loss, n}
+                              long long T, int S, int V) {
+  __shared__ float lds_m[8];
+  __shared__ float lds_s[8];
+  for (long long t = blockIdx.x; t < T; t += gridDim.x) {
+    const int s_pos = (int)(t % S);
+    long long label = -100;
+    if (s_pos < S - 1) label = labels[t + 1];
+    if (label < 0) {
+      if (threadIdx.x == 0) lse[t] = 0.0f;
+      continue;
+    }
+    const u16* row = logits + t * (long long)V;
+    // online logsumexp over the row, vec8
+    float m = -3.4e38f, sum = 0.0f;
+    const int nv = V / 8;
+    for (int c = threadIdx.x; c < nv; c += BLOCK) {
+      ushort4 a = reinterpret_cast<const ushort4*>(row)[2 * c];
+      ushort4 b = reinterpret_cast<const ushort4*>(row)[2 * c + 1];
+      float f[8] = {bf16_to_f32(a.x), bf16_to_f32(a.y), bf16_to_f32(a.z),
+                    bf16_to_f32(a.w), bf16_to_f32(b.x), bf16_to_f32(b.y),
+                    bf16_to_f32(b.z), bf16_to_f32(b.w)};
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float x = f[k];
+        if (x > m) { sum *= __expf(m - x); m = x; }
+        sum += __expf(x - m);
+      }
+    }
+    for (int tail = nv * 8 + threadIdx.x; tail < V; tail += BLOCK) {
+      float x = bf16_to_f32(row[tail]);
+      if (x > m) { sum *= __expf(m - x); m = x; }
+      sum += __expf(x - m);
+    }
+    // wave-reduce the (m, sum) pairs
+    for (int off = 32; off > 0; off >>= 1) {
+      float mo = __shfl_down(m, off, 64);
+      float so = __shfl_down(sum, off, 64);
+      float mn = fmaxf(m, mo);
+      sum = sum * __expf(m - mn) + so * __expf(mo - mn);
+      m = mn;
+    }
+    const int wave = threadIdx.x / 64;
+    if ((threadIdx.x & 63) == 0) { lds_m[wave] = m; lds_s[wave] = sum; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float M = lds_m[0], Ssum = lds_s[0];
+      for (int w2 = 1; w2 < BLOCK / 64; ++w2) {
+        float mo = lds_m[w2], so = lds_s[w2];
+        float mn = fmaxf(M, mo);
+        Ssum = Ssum * __expf(M - mn) + so * __expf(mo - mn);
+        M = mn;
+      }
+      const float l = M + __logf(Ssum);
+      lse[t] = l;
+      const float gold = bf16_to_f32(row[label]);
+      atomicAdd(acc, l - gold);
+      atomicAdd(acc + 1, 1.0f);
+    }
+    __syncthreads();
+  }
+}
+
+// dlogits[t, v] = scale * (exp(x - lse[t]) - 1[v == label])  (valid tokens)
+//              = 0                                           (otherwise)
+__global__ void ce_bwd_kernel(const u16* __restrict__ logits,
+                              const long long* __restrict__ labels,
+                              const float* __restrict__ lse,
+                              u16* __restrict__ dlogits,
+                              const float* __restrict__ acc,  // n_valid at [1]
+                              float dloss,                    // upstream grad
+                              long long T, int S, int V) {
+  const float scale = dloss / fmaxf(acc[1], 1.0f);
+  for (long long t = blockIdx.x; t < T; t += gridDim.x) {
+    const int s_pos = (int)(t % S);
+    long long label = -100;
+    if (s_pos < S - 1) label = labels[t + 1];
+    const u16* row = logits + t * (long long)V;
+    u16* drow = dlogits + t * (long long)V;
+    const int nv = V / 8;
+    if (label < 0) {
+      const ushort4 z = make_ushort4(0, 0, 0, 0);
+      for (int c = threadIdx.x; c < nv; c += BLOCK) {
+        reinterpret_cast<ushort4*>(drow)[2 * c] = z;
+        reinterpret_cast<ushort4*>(drow)[2 * c + 1] = z;
+      }
+      for (int tail = nv * 8 + threadIdx.x; tail < V; tail += BLOCK)
+        drow[tail] = 0;
+      continue;
+    }
+    const float l = lse[t];
+    for (int c = threadIdx.x; c < nv; c += BLOCK) {
+      ushort4 a = reinterpret_cast<const ushort4*>(row)[2 * c];
+      ushort4 b = reinterpret_cast<const ushort4*>(row)[2 * c + 1];
+      float f[8] = {bf16_to_f32(a.x), bf16_to_f32(a.y), bf16_to_f32(a.z),
+                    bf16_to_f32(a.w), bf16_to_f32(b.x), bf16_to_f32(b.y),
+                    bf16_to_f32(b.z), bf16_to_f32(b.w)};
+      u16 o[8];
+      const int base = c * 8;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float p = __expf(f[k] - l);
+        if ((long long)(base + k) == label) p -= 1.0f;
+        o[k] = f32_to_bf16(p * scale);
+      }
+      reinterpret_cast<ushort4*>(drow)[2 * c] = make_ushort4(o[0], o[1], o[2], o[3]);
+      reinterpret_cast<ushort4*>(drow)[2 * c + 1] = make_ushort4(o[4], o[5], o[6], o[7]);
+    }
+    for (int tail = nv * 8 + threadIdx.x; tail < V; tail += BLOCK) {
+      float p = __expf(bf16_to_f32(row[tail]) - l);
+      if ((long long)tail == label) p -= 1.0f;
+      drow[tail] = f32_to_bf16(p * scale);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void acco_ce_fwd(const void* logits, const long long* labels, float* lse,
+                 float* acc, long long T, int S, int V, hipStream_t s) {
+  int grid = (int)((T < 8192) ? T : 8192);
+  hipLaunchKernelGGL(ce_fwd_kernel, dim3(grid), dim3(BLOCK), 0, s,
+                     (const u16*)logits, labels, lse, acc, T, S, V);
+}
+
+void acco_ce_bwd(const void* logits, const long long* labels,
+                 const float* lse, void* dlogits, const float* acc,
+                 float dloss, long long T, int S, int V, hipStream_t s) {
+  int grid = (int)((T < 8192) ? T : 8192);
+  hipLaunchKernelGGL(ce_bwd_kernel, dim3(grid), dim3(BLOCK), 0, s,
+                     (const u16*)logits, labels, lse, (u16*)dlogits, acc,
+                     dloss, T, S, V);
+}
+
+}  // extern "C"

@@ -64,9 +64,12 @@ def _rotate_half(x: torch.Tensor) -> torch.Tensor:
 def rope_apply(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor,
                sin: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """Apply rotary embedding, HF-Llama rotate_half convention.
-    q,k: [B, H, S, D]; cos/sin: [S, D]."""
-    cos = cos.to(q.dtype)[None, None]
-    sin = sin.to(q.dtype)[None, None]
+
+    q: [B, S, H, D]; k: [B, S, Hkv, D] (the projection's natural contiguous
+    layout — the MI355X kernel walks D coalesced with no transpose copies);
+    cos/sin: [S, D]."""
+    cos = cos.to(q.dtype)[None, :, None, :]
+    sin = sin.to(q.dtype)[None, :, None, :]
     q2 = q * cos + _rotate_half(q) * sin
     k2 = k * cos + _rotate_half(k) * sin
     return q2, k2
@@ -77,13 +80,17 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      window: Optional[int] = None) -> torch.Tensor:
     """Causal (optionally banded/local) attention.
 
-    q: [B, H, S, D]; k, v: [B, Hkv, S, D] (GQA: H a multiple of Hkv).
+    q: [B, S, H, D]; k, v: [B, S, Hkv, D] (GQA: H a multiple of Hkv) —
+    the projections' natural contiguous layout. Returns [B, S, H, D].
     `scale=None` → 1/sqrt(D); GPT-Neo passes scale=1.0 (it does not scale).
     `window` (GPT-Neo local layers, window_size=256): query i attends to
     keys j with i-window < j <= i.
     """
-    B, H, S, D = q.shape
-    Hkv = k.shape[1]
+    B, S, H, D = q.shape
+    Hkv = k.shape[2]
+    q = q.transpose(1, 2)
+    k = k.transpose(1, 2)
+    v = v.transpose(1, 2)
     if Hkv != H:
         rep = H // Hkv
         k = k.repeat_interleave(rep, dim=1)
@@ -97,7 +104,8 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         mask = mask | (idx[None, :] <= idx[:, None] - window)
     scores = scores.masked_fill(mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
-    return torch.matmul(probs, v.float()).to(q.dtype)
+    out = torch.matmul(probs, v.float()).to(q.dtype)
+    return out.transpose(1, 2)
 
 
 def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor,

@@ -1,0 +1,135 @@
+"""Numerics of every gfx950 HIP kernel vs the fp32 torch reference
+(per-kernel parity tests). Tolerances account for bf16 I/O
+(~8 bit mantissa) with fp32 internal math on both sides."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _close_bf16(a, b, atol=3e-2, rtol=3e-2):
+    return torch.allclose(a.float(), b.float(), atol=atol, rtol=rtol)
+
+
+def setup_module(module):
+    torch.manual_seed(0)
+
+
+def test_swiglu_fwd_bwd():
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    g = torch.randn(4, 128, 512, device="cuda").bfloat16().requires_grad_(True)
+    u = torch.randn(4, 128, 512, device="cuda").bfloat16().requires_grad_(True)
+    out = ops.swiglu(g, u)
+    ref = torch_ref.swiglu(g.detach().float(), u.detach().float())
+    assert _close_bf16(out, ref)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    g32 = g.detach().float().requires_grad_(True)
+    u32 = u.detach().float().requires_grad_(True)
+    torch_ref.swiglu(g32, u32).backward(dout.float())
+    assert _close_bf16(g.grad, g32.grad)
+    assert _close_bf16(u.grad, u32.grad)
+
+
+def test_gelu_new_fwd_bwd():
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    x = torch.randn(2, 64, 768, device="cuda").bfloat16().requires_grad_(True)
+    out = ops.gelu_new(x)
+    ref = torch_ref.gelu_new(x.detach().float())
+    assert _close_bf16(out, ref)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    x32 = x.detach().float().requires_grad_(True)
+    torch_ref.gelu_new(x32).backward(dout.float())
+    assert _close_bf16(x.grad, x32.grad)
+
+
+@pytest.mark.parametrize("D", [768, 2048])
+def test_rmsnorm_fwd_bwd(D):
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    x = torch.randn(3, 37, D, device="cuda").bfloat16().requires_grad_(True)
+    w = (torch.randn(D, device="cuda") * 0.1 + 1.0).bfloat16().requires_grad_(True)
+    out = ops.rms_norm(x, w, 1e-5)
+    ref = torch_ref.rms_norm(x.detach().float(), w.detach().float(), 1e-5)
+    assert _close_bf16(out, ref)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    torch_ref.rms_norm(x32, w32, 1e-5).backward(dout.float())
+    assert _close_bf16(x.grad, x32.grad)
+    assert _close_bf16(w.grad, w32.grad, atol=0.1, rtol=0.05)
+
+
+def test_layernorm_fwd_bwd():
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    D = 768
+    x = torch.randn(4, 32, D, device="cuda").bfloat16().requires_grad_(True)
+    w = (torch.randn(D, device="cuda") * 0.1 + 1.0).bfloat16().requires_grad_(True)
+    b = (torch.randn(D, device="cuda") * 0.1).bfloat16().requires_grad_(True)
+    out = ops.layer_norm(x, w, b, 1e-5)
+    ref = torch_ref.layer_norm(x.detach().float(), w.detach().float(),
+                               b.detach().float(), 1e-5)
+    assert _close_bf16(out, ref)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    b32 = b.detach().float().requires_grad_(True)
+    torch_ref.layer_norm(x32, w32, b32, 1e-5).backward(dout.float())
+    assert _close_bf16(x.grad, x32.grad)
+    assert _close_bf16(w.grad, w32.grad, atol=0.1, rtol=0.05)
+    assert _close_bf16(b.grad, b32.grad, atol=0.1, rtol=0.05)
+
+
+@pytest.mark.parametrize("D", [64, 128])
+def test_rope_fwd_bwd(D):
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    B, S, H = 2, 96, 4
+    q = torch.randn(B, S, H, D, device="cuda").bfloat16().requires_grad_(True)
+    k = torch.randn(B, S, 2, D, device="cuda").bfloat16().requires_grad_(True)
+    cos, sin = torch_ref.rope_cos_sin(S, D, 10000.0, "cuda")
+    q2, k2 = ops.rope_apply(q, k, cos, sin)
+    q2r, k2r = torch_ref.rope_apply(q.detach().float(), k.detach().float(),
+                                    cos, sin)
+    assert _close_bf16(q2, q2r)
+    assert _close_bf16(k2, k2r)
+    dq = torch.randn_like(q2)
+    dk = torch.randn_like(k2)
+    (q2 * dq.detach()).sum().backward(retain_graph=True)
+    # grad check via autograd on the fp32 reference
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    q2f, k2f = torch_ref.rope_apply(q32, k32, cos, sin)
+    (q2f * dq.detach().float()).sum().backward()
+    assert _close_bf16(q.grad, q32.grad)
+
+
+def test_ce_fwd_bwd():
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    B, S, V = 2, 33, 1031          # odd V exercises the tail path
+    logits = (torch.randn(B, S, V, device="cuda") * 2).bfloat16().requires_grad_(True)
+    labels = torch.randint(0, V, (B, S), device="cuda")
+    labels[0, 5] = -100            # exercised only through the shift
+    loss = ops.causal_lm_loss(logits, labels)
+    ref = torch_ref.causal_lm_loss(logits.detach().float(), labels)
+    assert torch.allclose(loss, ref, atol=2e-3, rtol=2e-3), (loss, ref)
+    loss.backward()
+    l32 = logits.detach().float().requires_grad_(True)
+    torch_ref.causal_lm_loss(l32, labels).backward()
+    assert _close_bf16(logits.grad, l32.grad, atol=1e-3, rtol=5e-2)
+
+
+def test_model_ops_route_to_hip():
+    """The live Llama/GPT-Neo blocks must be running HIP kernels, not ATen."""
+    from acco_amd import ops
+    for k in ["swiglu_fwd", "gelu_fwd", "rmsnorm_fwd", "layernorm_fwd",
+              "rope_fwd", "ce_fwd", "fused_adamw"]:
+        assert ops.have_kernel(k), k
