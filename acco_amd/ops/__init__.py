@@ -113,6 +113,11 @@ def causal_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      window: Optional[int] = None) -> torch.Tensor:
     if _use_ref(q, "attn_fwd"):
         return torch_ref.causal_attention(q, k, v, scale=scale, window=window)
+    S, D = q.shape[1], q.shape[3]
+    if S % 64 != 0 or D not in (64, 128) or q.dtype != torch.bfloat16:
+        # shapes outside the flash kernel's contract (rare: tests, odd
+        # eval batches) run the reference math
+        return torch_ref.causal_attention(q, k, v, scale=scale, window=window)
     from acco_amd.ops.autograd import AttentionFn
     return AttentionFn.apply(q, k, v, scale, window)
 

@@ -117,10 +117,40 @@ class CausalLMLossFn(torch.autograd.Function):
 
 
 class AttentionFn(torch.autograd.Function):
-    """Flash-style causal attention (gfx950 MFMA) — kernel lands next;
-    dispatch guards on have_kernel('attn_fwd') so this is unreachable
-    until then."""
+    """Flash-style causal attention, gfx950 MFMA (fwd: online softmax;
+    bwd: FA2-style recompute, dq + dkv kernels). [B,S,H,D] layout,
+    D ∈ {64,128}, S % 64 == 0 (the dispatch falls back to torch_ref for
+    other shapes)."""
 
     @staticmethod
     def forward(ctx, q, k, v, scale, window):
-        raise NotImplementedError("attn_fwd kernel not built yet")
+        import math
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        w = int(window) if window else 0
+        q = q.contiguous()
+        k = k.contiguous()
+        v = v.contiguous()
+        o, lse = ops.hip_ext().attn_fwd(q, k, v, float(scale), w)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = float(scale)
+        ctx.window = w
+        return o
+
+    @staticmethod
+    def backward(ctx, dO):
+        q, k, v, o, lse = ctx.saved_tensors
+        dO = dO.contiguous()
+        # Delta[b,h,s] = rowsum(dO ∘ O) in fp32, laid out [B,H,S]
+        delta = (dO.float() * o.float()).sum(-1).permute(0, 2, 1).contiguous()
+        dq, dkq, dvq = ops.hip_ext().attn_bwd(q, k, v, dO, lse, delta,
+                                              ctx.scale, ctx.window)
+        B, S, H, D = q.shape
+        Hkv = k.shape[2]
+        if Hkv != H:
+            rep = H // Hkv
+            dk = dkq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32).to(k.dtype)
+            dv = dvq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32).to(v.dtype)
+        else:
+            dk, dv = dkq, dvq
+        return dq, dk, dv, None, None

@@ -1,0 +1,363 @@
+// Flash-style causal attention BACKWARD for gfx950 (CDNA4 MFMA) —
+// SURVEY.md §2.5 K2 for the attention block, §7 "hard parts".
+//
+// FlashAttention-2 style two-kernel recompute scheme (no S×S
+// materialization, no atomics):
+//   dq kernel : grid over Q tiles; for each kv tile j<=qt recompute
+//               P = exp(S - lse), dP = dO·V^T, dS = P∘(dP - Delta),
+//               dQ += dS·K.
+//   dkv kernel: grid over KV tiles; for each q tile qt>=j accumulate
+//               dV += P^T·dO and dK += dS^T·Q.
+//   Delta[b,h,q] = rowsum(dO ∘ O) is computed by the Python wrapper
+//   (one fused multiply-reduce).
+//
+// Same fragment-layout playbook as the forward: swapped/non-swapped MFMA
+// picked so every global A/B fragment is a contiguous 16-byte per-lane
+// load; C-layout products that must feed the *A* side of the next MFMA
+// (P, dS) take one per-wave LDS round trip; operands needed transposed on
+// the *B* side (K^T, Q^T, dO^T) are staged transposed in LDS once per tile
+// and shared by the 4 waves.
+
+#include "common.h"
+
+namespace {
+
+using u16 = unsigned short;
+using short8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int TILE = 64;
+constexpr int PAD = 8;
+constexpr int LST = TILE + PAD;       // LDS row stride (bf16 elems)
+
+#define MFMA(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
+
+// stage a [TILE × D] global tile TRANSPOSED into LDS [D][LST]
+template <int D>
+ACCO_DEV void stage_transposed(const u16* src, long long row_stride,
+                               u16* dst_lds) {
+  const int r = threadIdx.x & 63;                   // tile row
+  for (int dg = threadIdx.x >> 6; dg < D / 16; dg += 4) {
+    const u16* p = src + (long long)r * row_stride + dg * 16;
+    ushort4 a = reinterpret_cast<const ushort4*>(p)[0];
+    ushort4 b = reinterpret_cast<const ushort4*>(p)[1];
+    ushort4 c = reinterpret_cast<const ushort4*>(p)[2];
+    ushort4 d = reinterpret_cast<const ushort4*>(p)[3];
+    u16 vals[16] = {a.x, a.y, a.z, a.w, b.x, b.y, b.z, b.w,
+                    c.x, c.y, c.z, c.w, d.x, d.y, d.z, d.w};
+#pragma unroll
+    for (int i = 0; i < 16; ++i)
+      dst_lds[(dg * 16 + i) * LST + r] = vals[i];
+  }
+}
+
+// ------------------------------------------------------------------- dQ
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
+                        const u16* __restrict__ v, const u16* __restrict__ dO,
+                        const float* __restrict__ lse,
+                        const float* __restrict__ delta,   // [B,H,S]
+                        u16* __restrict__ dq,
+                        int S, int H, int Hkv, float scale, int window) {
+  constexpr int KS = D / 32;
+  constexpr int DT = D / 16;
+  const int qt = blockIdx.x, bh = blockIdx.y;
+  const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int lg = lane >> 4, lc = lane & 15;
+
+  extern __shared__ __attribute__((aligned(16))) u16 smem[];
+  u16* kT_lds = smem;                            // [D][LST] K^T tile
+  u16* ds_lds = smem + D * LST + wave * 16 * LST;  // per-wave dS tile
+
+  const long long qs = (long long)H * D, ks = (long long)Hkv * D;
+  const int q0 = qt * TILE + wave * 16;
+  const u16* Qp = q + ((long long)b * S + q0) * qs + (long long)h * D;
+  const u16* dOp = dO + ((long long)b * S + q0) * qs + (long long)h * D;
+  const u16* Kb = k + (long long)b * S * ks + (long long)hkv * D;
+  const u16* Vb = v + (long long)b * S * ks + (long long)hkv * D;
+
+  // per-lane fragments held for the whole kernel:
+  // Q (B of swapped S^T), dO (B of swapped dP^T): lane = [row lc][d lg*8+]
+  short8 qf[KS], dof[KS];
+#pragma unroll
+  for (int s = 0; s < KS; ++s) {
+    qf[s] = *reinterpret_cast<const short8*>(Qp + (long long)lc * qs + s * 32 + lg * 8);
+    dof[s] = *reinterpret_cast<const short8*>(dOp + (long long)lc * qs + s * 32 + lg * 8);
+  }
+  // per-col (q=lc) stats
+  const float lse_c = lse[(long long)bh * S + q0 + lc];
+  const float delta_c = delta[(long long)bh * S + q0 + lc];
+
+  f32x4 acc_dq[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t) acc_dq[t] = {0.f, 0.f, 0.f, 0.f};
+
+  int j_lo = 0;
+  if (window > 0) {
+    int kv_min = qt * TILE - window + 1;
+    if (kv_min > 0) j_lo = kv_min / TILE;
+  }
+
+  for (int j = j_lo; j <= qt; ++j) {
+    __syncthreads();
+    stage_transposed<D>(Kb + (long long)(j * TILE) * ks, ks, kT_lds);
+    __syncthreads();
+
+    const u16* Kt = Kb + (long long)(j * TILE) * ks;
+    const u16* Vt = Vb + (long long)(j * TILE) * ks;
+    // S^T = K·Q^T and dP^T = V·dO^T (both C: col=q=lc, row=kv spread)
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int m16 = 0; m16 < 4; ++m16) {
+      f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int s = 0; s < KS; ++s) {
+        short8 kf = *reinterpret_cast<const short8*>(
+            Kt + (long long)(m16 * 16 + lc) * ks + s * 32 + lg * 8);
+        short8 vf = *reinterpret_cast<const short8*>(
+            Vt + (long long)(m16 * 16 + lc) * ks + s * 32 + lg * 8);
+        a1 = MFMA(kf, qf[s], a1);
+        a2 = MFMA(vf, dof[s], a2);
+      }
+      st[m16] = a1;
+      dpt[m16] = a2;
+    }
+
+    // dS^T = P^T ∘ (dP^T - Delta) * scale, P = exp(S*scale - lse)
+    const int q_g = qt * TILE + wave * 16 + lc;
+#pragma unroll
+    for (int m16 = 0; m16 < 4; ++m16) {
+      u16 pk[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kv_g = j * TILE + m16 * 16 + lg * 4 + r;
+        bool valid = (kv_g <= q_g);
+        if (window > 0) valid = valid && (kv_g > q_g - window);
+        const float pval = valid ? __expf(st[m16][r] * scale - lse_c) : 0.0f;
+        const float dsv = pval * (dpt[m16][r] - delta_c) * scale;
+        pk[r] = f32_to_bf16(dsv);
+      }
+      *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
+          make_ushort4(pk[0], pk[1], pk[2], pk[3]);
+    }
+    __builtin_amdgcn_s_waitcnt(0);   // lgkm: own-wave LDS writes
+
+    // dQ += dS·K : A = dS[q=lc][kv], B = K^T[kv][d] from kT_lds
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      short8 dsa = *reinterpret_cast<const short8*>(
+          ds_lds + lc * LST + s * 32 + lg * 8);
+#pragma unroll
+      for (int t = 0; t < DT; ++t) {
+        // B[k=kv][col=d]: lane reads kT_lds[d=t*16+lc][kv=s*32+lg*8 ..]
+        short8 kb = *reinterpret_cast<const short8*>(
+            kT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+        acc_dq[t] = MFMA(dsa, kb, acc_dq[t]);
+      }
+    }
+  }
+
+  // store dQ rows q=lg*4+r, col d=t*16+lc
+  u16* dQp = dq + ((long long)b * S + q0) * qs + (long long)h * D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+#pragma unroll
+    for (int t = 0; t < DT; ++t)
+      dQp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
+          f32_to_bf16(acc_dq[t][r]);
+}
+
+// ---------------------------------------------------------------- dK, dV
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
+                         const u16* __restrict__ v, const u16* __restrict__ dO,
+                         const float* __restrict__ lse,
+                         const float* __restrict__ delta,
+                         u16* __restrict__ dk, u16* __restrict__ dv,
+                         int S, int H, int Hkv, float scale, int window,
+                         int kv_repeat) {
+  // one workgroup per (kv tile, b, h-of-Q); grads for the SHARED kv head
+  // are accumulated over its kv_repeat query heads by h-loop inside.
+  constexpr int KS = D / 32;
+  constexpr int DT = D / 16;
+  const int j = blockIdx.x, bh = blockIdx.y;
+  const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int lg = lane >> 4, lc = lane & 15;
+
+  extern __shared__ __attribute__((aligned(16))) u16 smem[];
+  u16* qT_lds = smem;                               // [D][LST]
+  u16* doT_lds = smem + D * LST;                    // [D][LST]
+  u16* p_lds = smem + 2 * D * LST + wave * 16 * LST;      // per-wave P^T
+  u16* ds_lds = smem + 2 * D * LST + (4 + wave) * 16 * LST;  // per-wave dS^T
+
+  const long long qs = (long long)H * D, ks = (long long)Hkv * D;
+  const int kv0 = j * TILE + wave * 16;             // wave's 16 kv rows
+  const u16* Kp = k + ((long long)b * S + kv0) * ks + (long long)hkv * D;
+  const u16* Vp = v + ((long long)b * S + kv0) * ks + (long long)hkv * D;
+  const u16* Qb = q + (long long)b * S * qs + (long long)h * D;
+  const u16* dOb = dO + (long long)b * S * qs + (long long)h * D;
+
+  // K, V fragments as B operands (transposed reads are contiguous):
+  // B[k=d][col=kv]: lane = K[kv=lc][d=lg*8+i]
+  short8 kTf[KS], vTf[KS];
+#pragma unroll
+  for (int s = 0; s < KS; ++s) {
+    kTf[s] = *reinterpret_cast<const short8*>(Kp + (long long)lc * ks + s * 32 + lg * 8);
+    vTf[s] = *reinterpret_cast<const short8*>(Vp + (long long)lc * ks + s * 32 + lg * 8);
+  }
+
+  f32x4 acc_dk[DT], acc_dv[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t) {
+    acc_dk[t] = {0.f, 0.f, 0.f, 0.f};
+    acc_dv[t] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  int qt_hi = S / TILE - 1;
+  if (window > 0) {
+    // largest q tile with q - window < kv_max  →  q < kv_max + window
+    const int q_lim = j * TILE + TILE - 1 + window;       // exclusive-ish
+    qt_hi = min(qt_hi, q_lim / TILE);
+  }
+
+  for (int qt = j; qt <= qt_hi; ++qt) {
+    __syncthreads();
+    stage_transposed<D>(Qb + (long long)(qt * TILE) * qs, qs, qT_lds);
+    stage_transposed<D>(dOb + (long long)(qt * TILE) * qs, qs, doT_lds);
+    __syncthreads();
+
+    const u16* Qt = Qb + (long long)(qt * TILE) * qs;
+    const u16* dOt = dOb + (long long)(qt * TILE) * qs;
+
+    // S = Q·K^T, dP = dO·V^T (C: col=kv=lc, row=q spread)
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int m16 = 0; m16 < 4; ++m16) {
+      f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int s = 0; s < KS; ++s) {
+        short8 qfr = *reinterpret_cast<const short8*>(
+            Qt + (long long)(m16 * 16 + lc) * qs + s * 32 + lg * 8);
+        short8 dofr = *reinterpret_cast<const short8*>(
+            dOt + (long long)(m16 * 16 + lc) * qs + s * 32 + lg * 8);
+        a1 = MFMA(qfr, kTf[s], a1);    // wait: A rows must be q…
+        a2 = MFMA(dofr, vTf[s], a2);
+      }
+      st[m16] = a1;
+      dpt[m16] = a2;
+    }
+    // NOTE on layout: A = Q fragment has rows q = m16*16+lc?  No:
+    // A[row][k] with row=lane&15 — the m16 loop walks q sub-tiles, so C of
+    // sub-tile m16 covers q rows m16*16..+15 with col=kv=lc … but C's col
+    // belongs to B (K^T cols = the wave's 16 kv rows).  See python-side
+    // layout test; the equations below use: C[m16][r]: q_g = qt*TILE +
+    // m16*16 + lg*4 + r, kv_g = kv0 + lc.
+
+    const int kv_g = kv0 + lc;
+#pragma unroll
+    for (int m16 = 0; m16 < 4; ++m16) {
+      u16 ppk[4], dsk[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q_g = qt * TILE + m16 * 16 + lg * 4 + r;
+        bool valid = (kv_g <= q_g) && (q_g < S);
+        if (window > 0) valid = valid && (kv_g > q_g - window);
+        const float lse_q = lse[(long long)bh * S + min(q_g, S - 1)];
+        const float del_q = delta[(long long)bh * S + min(q_g, S - 1)];
+        const float pval = valid ? __expf(st[m16][r] * scale - lse_q) : 0.0f;
+        const float dsv = pval * (dpt[m16][r] - del_q) * scale;
+        ppk[r] = f32_to_bf16(pval);
+        dsk[r] = f32_to_bf16(dsv);
+      }
+      // store P^T and dS^T: row kv=lc, col q = m16*16 + lg*4 + r
+      *reinterpret_cast<ushort4*>(p_lds + lc * LST + m16 * 16 + lg * 4) =
+          make_ushort4(ppk[0], ppk[1], ppk[2], ppk[3]);
+      *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
+          make_ushort4(dsk[0], dsk[1], dsk[2], dsk[3]);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // dV += P^T·dO : A = P^T[kv=lc][q], B = dO^T[q][d] from doT_lds
+    // dK += dS^T·Q : A = dS^T[kv=lc][q], B = Q^T[q][d] from qT_lds
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      short8 pa = *reinterpret_cast<const short8*>(
+          p_lds + lc * LST + s * 32 + lg * 8);
+      short8 dsa = *reinterpret_cast<const short8*>(
+          ds_lds + lc * LST + s * 32 + lg * 8);
+#pragma unroll
+      for (int t = 0; t < DT; ++t) {
+        short8 dob = *reinterpret_cast<const short8*>(
+            doT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+        short8 qb = *reinterpret_cast<const short8*>(
+            qT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+        acc_dv[t] = MFMA(pa, dob, acc_dv[t]);
+        acc_dk[t] = MFMA(dsa, qb, acc_dk[t]);
+      }
+    }
+  }
+
+  // store dK/dV rows kv = kv0 + lg*4 + r, col d = t*16+lc.
+  // GQA (kv_repeat > 1): several query heads share this kv head → the
+  // wrapper allocates per-QUERY-head buffers [B,S,H,D] and reduces over
+  // the group afterwards, so each workgroup writes its own h slice.
+  u16* dKp = dk + ((long long)b * S + kv0) * qs + (long long)h * D;
+  u16* dVp = dv + ((long long)b * S + kv0) * qs + (long long)h * D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+#pragma unroll
+    for (int t = 0; t < DT; ++t) {
+      dKp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
+          f32_to_bf16(acc_dk[t][r]);
+      dVp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
+          f32_to_bf16(acc_dv[t][r]);
+    }
+  (void)kv_repeat;
+}
+
+}  // namespace
+
+extern "C" {
+
+void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
+                      const void* dO, const float* lse, const float* delta,
+                      void* dq, int B, int S, int H, int Hkv, int D,
+                      float scale, int window, hipStream_t stream) {
+  dim3 grid(S / TILE, B * H);
+  const int lds = (D + 4 * 16) * LST * sizeof(u16);
+  if (D == 64)
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<64>, grid, dim3(256), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
+                       scale, window);
+  else
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<128>, grid, dim3(256), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
+                       scale, window);
+}
+
+void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
+                       const void* dO, const float* lse, const float* delta,
+                       void* dk, void* dv, int B, int S, int H, int Hkv,
+                       int D, float scale, int window, hipStream_t stream) {
+  dim3 grid(S / TILE, B * H);
+  const int lds = (2 * D + 2 * 4 * 16) * LST * sizeof(u16);
+  const int rep = H / Hkv;
+  if (D == 64)
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel<64>, grid, dim3(256), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
+                       Hkv, scale, window, rep);
+  else
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel<128>, grid, dim3(256), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
+                       Hkv, scale, window, rep);
+}
+
+}  // extern "C"

@@ -35,6 +35,17 @@ extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
 extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
                             void*, const float*, float, long long, int, int,
                             hipStream_t);
+extern "C" void acco_attn_fwd(const void*, const void*, const void*, void*,
+                              float*, int, int, int, int, int, float, int,
+                              hipStream_t);
+extern "C" void acco_attn_bwd_dq(const void*, const void*, const void*,
+                                 const void*, const float*, const float*,
+                                 void*, int, int, int, int, int, float, int,
+                                 hipStream_t);
+extern "C" void acco_attn_bwd_dkv(const void*, const void*, const void*,
+                                  const void*, const float*, const float*,
+                                  void*, void*, int, int, int, int, int,
+                                  float, int, hipStream_t);
 
 namespace {
 
@@ -215,6 +226,48 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
   return dlogits;
 }
 
+// ---- flash attention ([B, S, H, D] layout, D in {64, 128}, S % 64 == 0)
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 double scale, int64_t window) {
+  CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
+  const int B = (int)q.size(0), S = (int)q.size(1), H = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  TORCH_CHECK(S % 64 == 0 && (D == 64 || D == 128) && H % Hkv == 0);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  acco_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                lse.data_ptr<float>(), B, S, H, Hkv, D, (float)scale,
+                (int)window, cur_stream());
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor dO, at::Tensor lse,
+                                 at::Tensor delta, double scale,
+                                 int64_t window) {
+  CHECK_BF16_CONTIG(q); CHECK_BF16_CONTIG(k); CHECK_BF16_CONTIG(v);
+  CHECK_BF16_CONTIG(dO);
+  const int B = (int)q.size(0), S = (int)q.size(1), H = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  TORCH_CHECK(delta.scalar_type() == at::kFloat && delta.is_contiguous());
+  auto dq = at::empty_like(q);
+  // dk/dv are emitted per QUERY head; the Python wrapper group-reduces
+  auto dk = at::empty_like(q);
+  auto dv = at::empty_like(q);
+  acco_attn_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(), dO.data_ptr(),
+                   lse.data_ptr<float>(), delta.data_ptr<float>(),
+                   dq.data_ptr(), B, S, H, Hkv, D, (float)scale, (int)window,
+                   cur_stream());
+  acco_attn_bwd_dkv(q.data_ptr(), k.data_ptr(), v.data_ptr(), dO.data_ptr(),
+                    lse.data_ptr<float>(), delta.data_ptr<float>(),
+                    dk.data_ptr(), dv.data_ptr(), B, S, H, Hkv, D,
+                    (float)scale, (int)window, cur_stream());
+  return {dq, dk, dv};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -232,5 +285,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.attr("_gfx950") = true;
 }
