@@ -37,6 +37,7 @@ from acco_amd.engine.sharded_adamw import ShardedAdamW
 from acco_amd.engine.scheduler import LRSchedule
 from acco_amd.engine.streams import Event, Stream
 from acco_amd.parallel.comm import CommBackend, ShardSpec
+from acco_amd.utils.profiling import trace_range
 
 
 class AccoEngine:
@@ -108,7 +109,8 @@ class AccoEngine:
     def gradient_step(self, inputs: Dict) -> None:
         """One micro-batch forward/backward accumulating into grads arena
         (reference gradient_step :18-39)."""
-        loss = self.forward_backward(inputs)
+        with trace_range("acco/gradient_step"):
+            loss = self.forward_backward(inputs)
         self.count_grad_local += 1
         self.micro_steps += 1
         self.loss_static.copy_(loss.detach().float().reshape(1))
@@ -138,6 +140,11 @@ class AccoEngine:
         count of the round."""
         if advance_sched is None:
             advance_sched = commit
+        with trace_range("acco/communication_round"):
+            return self._communication_round(commit, advance_sched)
+
+    @torch.no_grad()
+    def _communication_round(self, commit: bool, advance_sched: bool) -> int:
         work_count = self.comm.all_reduce_sum_async(self.count_grad_this_round)
         rs = [self.comm.reduce_scatter_bucket_async(self.com_buffer, self.spec,
                                                     j, self.rank)

@@ -204,12 +204,22 @@ class DecoupledTrainer:
     def forward_backward(self, inputs: Dict) -> torch.Tensor:
         """(reference gradient_step :18-39; loss divided by n_acc only in
         DDP mode — ACCO averages by global grad count instead)"""
+        ls = float(getattr(self.args, "label_smoothing_factor", 0) or 0)
         with self._autocast():
-            if "labels" in inputs:
-                outputs = self.model(**inputs)
+            if ls != 0.0:
+                # label-smoothed path (reference compute_loss :262-282 with
+                # LabelSmoother): loss computed outside the model head
+                from acco_amd.ops import torch_ref
+                labels = inputs.get("labels", inputs["input_ids"])
+                logits = self.model(inputs["input_ids"])[0]
+                loss = torch_ref.label_smoothed_causal_lm_loss(
+                    logits, labels, ls) / self.loss_div
             else:
-                outputs = self.model(**inputs, labels=inputs["input_ids"])
-            loss = outputs[0] / self.loss_div
+                if "labels" in inputs:
+                    outputs = self.model(**inputs)
+                else:
+                    outputs = self.model(**inputs, labels=inputs["input_ids"])
+                loss = outputs[0] / self.loss_div
         loss.backward()
         return loss.detach() * self.loss_div
 

@@ -22,6 +22,7 @@ extern "C" void acco_rmsnorm_fwd(const void*, const void*, void*, void*,
 extern "C" void acco_rmsnorm_bwd(const void*, const void*, const void*,
                                  const void*, void*, void*, long long, int,
                                  hipStream_t);
+extern "C" int acco_norm_bwd_grid(long long R);
 extern "C" void acco_layernorm_fwd(const void*, const void*, const void*,
                                    void*, void*, void*, long long, int, float,
                                    hipStream_t);
@@ -141,11 +142,12 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
+  const int grid = acco_norm_bwd_grid(R);
+  auto dw_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
-                   rstd.data_ptr(), dx.data_ptr(), dw.data_ptr(), R, D,
+                   rstd.data_ptr(), dx.data_ptr(), dw_part.data_ptr(), R, D,
                    cur_stream());
-  return {dx, dw};
+  return {dx, dw_part.sum(0)};
 }
 
 // ---- LayerNorm
@@ -171,12 +173,14 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
   auto dx = at::empty_like(x);
-  auto dw = at::zeros({D}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({D}, x.options().dtype(at::kFloat));
+  const int grid = acco_norm_bwd_grid(R);
+  auto dw_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
+  auto db_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
-                     dw.data_ptr(), db.data_ptr(), R, D, cur_stream());
-  return {dx, dw, db};
+                     dw_part.data_ptr(), db_part.data_ptr(), R, D,
+                     cur_stream());
+  return {dx, dw_part.sum(0), db_part.sum(0)};
 }
 
 // ---- RoPE ([B, S, H, D] contiguous)

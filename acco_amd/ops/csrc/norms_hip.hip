@@ -107,7 +107,7 @@ __global__ __launch_bounds__(BLOCK)
 void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                         const u16* __restrict__ w,
                         const float* __restrict__ rstd, u16* __restrict__ dx,
-                        float* __restrict__ dw,   // fp32, zeroed
+                        float* __restrict__ dw_part,  // [gridDim.x, D] fp32
                         long long R, int D) {
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];
@@ -158,12 +158,16 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
         store8(dxr + cid[j] * VEC, o);
       }
   }
+  // one coalesced partial row per block (atomicAdd on 2k fp32 addresses
+  // from 4k blocks serialized ~3x worse than the pre-fix kernel; the
+  // wrapper reduces the [grid, D] scratch with one tiny torch sum)
+  float* out_row = dw_part + (long long)blockIdx.x * D;
 #pragma unroll
   for (int j = 0; j < CH; ++j)
     if (act[j])
 #pragma unroll
       for (int kk = 0; kk < VEC; ++kk)
-        atomicAdd(dw + cid[j] * VEC + kk, dwacc[j][kk]);
+        out_row[cid[j] * VEC + kk] = dwacc[j][kk];
 }
 
 // ---------------------------------------------------------- LayerNorm fwd
@@ -245,8 +249,10 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                           const u16* __restrict__ w,
                           const float* __restrict__ mean_in,
                           const float* __restrict__ rstd_in,
-                          u16* __restrict__ dx, float* __restrict__ dw,
-                          float* __restrict__ db, long long R, int D) {
+                          u16* __restrict__ dx,
+                          float* __restrict__ dw_part,  // [grid, D]
+                          float* __restrict__ db_part,  // [grid, D]
+                          long long R, int D) {
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];
   const int nv = D / VEC;
@@ -304,13 +310,15 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
         store8(dxr + cid[j] * VEC, o);
       }
   }
+  float* wrow = dw_part + (long long)blockIdx.x * D;
+  float* brow = db_part + (long long)blockIdx.x * D;
 #pragma unroll
   for (int j = 0; j < CH; ++j)
     if (act[j])
 #pragma unroll
       for (int kk = 0; kk < VEC; ++kk) {
-        atomicAdd(dw + cid[j] * VEC + kk, dwacc[j][kk]);
-        atomicAdd(db + cid[j] * VEC + kk, dbacc[j][kk]);
+        wrow[cid[j] * VEC + kk] = dwacc[j][kk];
+        brow[cid[j] * VEC + kk] = dbacc[j][kk];
       }
 }
 
@@ -329,6 +337,8 @@ int chunks_for(int D) {
 
 extern "C" {
 
+int acco_norm_bwd_grid(long long R) { return (int)((R < 2048) ? R : 2048); }
+
 void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
                       long long R, int D, float eps, hipStream_t s) {
   int grid = (int)((R < 8192) ? R : 8192);
@@ -346,7 +356,7 @@ void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
 void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
                       const void* rstd, void* dx, void* dw_fp32,
                       long long R, int D, hipStream_t s) {
-  int grid = (int)((R < 4096) ? R : 4096);
+  int grid = acco_norm_bwd_grid(R);
   const int lds = D * sizeof(u16);
 #define L(CH) hipLaunchKernelGGL(rmsnorm_bwd_kernel<CH>, dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, (const u16*)w, \
@@ -377,7 +387,7 @@ void acco_layernorm_bwd(const void* dy, const void* x, const void* w,
                         const void* mean, const void* rstd, void* dx,
                         void* dw_fp32, void* db_fp32, long long R, int D,
                         hipStream_t s) {
-  int grid = (int)((R < 4096) ? R : 4096);
+  int grid = acco_norm_bwd_grid(R);
   const int lds = D * sizeof(u16);
 #define L(CH) hipLaunchKernelGGL(layernorm_bwd_kernel<CH>, dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, (const u16*)w, \

@@ -117,6 +117,26 @@ def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor,
                            shift_labels.view(-1), ignore_index=ignore_index)
 
 
+def label_smoothed_causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor,
+                                  epsilon: float,
+                                  ignore_index: int = -100) -> torch.Tensor:
+    """HF LabelSmoother-equivalent shifted loss
+    (reference utils/trainer_utils.py:862-902, used when
+    label_smoothing_factor != 0 — trainer_base.py:63-68):
+    (1-eps)·NLL + eps·(mean over vocab of -log p), masked mean."""
+    shift_logits = logits[..., :-1, :].contiguous().float()
+    shift_labels = labels[..., 1:].contiguous()
+    logp = F.log_softmax(shift_logits, dim=-1)
+    mask = shift_labels.eq(ignore_index)
+    safe = shift_labels.clamp(min=0).unsqueeze(-1)
+    nll = -logp.gather(-1, safe).squeeze(-1).masked_fill(mask, 0.0)
+    smooth = -logp.sum(-1).masked_fill(mask, 0.0)
+    n = (~mask).sum().clamp(min=1)
+    nll = nll.sum() / n
+    smooth = smooth.sum() / (n * shift_logits.size(-1))
+    return (1.0 - epsilon) * nll + epsilon * smooth
+
+
 @torch.no_grad()
 def fused_adamw_step(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
                      v: torch.Tensor, step: int, lr: float, beta1: float,

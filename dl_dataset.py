@@ -1,0 +1,40 @@
+"""Offline dataset tokenizer (reference dl_dataset.py:8-34 capability):
+tokenize a HF dataset to const-length packed sequences and save to disk.
+
+    python dl_dataset.py data=openwebtext model=gptneo \
+        train.max_length=1024 out_dir=./tokenized/openwebtext
+"""
+
+from __future__ import annotations
+
+import sys
+
+from acco_amd.config import load_config
+from acco_amd.data.packing import make_tokenize_const_len_fn
+
+
+def main(argv=None):
+    cfg = load_config(argv if argv is not None else sys.argv[1:])
+    out_dir = cfg.get("out_dir", "./tokenized")
+
+    import datasets
+    from transformers import AutoTokenizer
+
+    ds = datasets.load_dataset(cfg.data.path)
+    ds = ds["train"].train_test_split(0.05, seed=42)   # reference main.py:50
+    tokenizer = AutoTokenizer.from_pretrained(cfg.model.tokenizer)
+    tokenizer.pad_token_id = tokenizer.eos_token_id
+
+    fn = make_tokenize_const_len_fn(tokenizer, "text", cfg.train.max_length)
+    out = {}
+    for split in ("train", "test"):
+        out[split] = ds[split].map(
+            fn, batched=True, remove_columns=ds[split].column_names,
+            num_proc=cfg.train.dataloader_num_workers or 1)
+    packed = datasets.DatasetDict(out)
+    packed.save_to_disk(out_dir)
+    print(f"saved packed dataset to {out_dir}")
+
+
+if __name__ == "__main__":
+    main()

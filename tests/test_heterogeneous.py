@@ -1,0 +1,69 @@
+"""Heterogeneous per-rank accumulation (BASELINE.json config 5): ranks run
+DIFFERENT n_grad_accumulation; the grad-count-weighted averaging
+(reference trainer_decoupled.py:85-98) must keep ranks in sync and match
+the oracle replay, which uses the recorded per-rank counts."""
+
+import os
+
+import torch
+import torch.nn as nn
+
+from tests.conftest import run_distributed
+from tests.dist_utils import init_worker, teardown_worker
+from tests.test_acco_oracle import D, TARGET, make_batches, oracle_replay
+
+
+def _worker_hetero(rank, world, port, tmpdir):
+    init_worker(rank, world, port)
+    from acco_amd.engine import arena
+    from acco_amd.engine.acco import AccoEngine
+    from acco_amd.engine.scheduler import LRSchedule
+    from acco_amd.engine.sharded_adamw import ShardedAdamW
+    from acco_amd.parallel.comm import CommBackend, ShardSpec
+
+    torch.manual_seed(7)
+    model = nn.Linear(D, 1, bias=False)
+    device = torch.device("cpu")
+    spec = ShardSpec.build(D, world, buckets=2, align=2)
+    params = arena.flatten_params(model, torch.float32, device, pad_to=spec.total)
+    grads = arena.attach_grad_arena(model, torch.float32, device, pad_to=spec.total)
+    comm = CommBackend(device)
+    opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
+    sched = LRSchedule(1e-2, 4, TARGET, "cosine")
+    it = iter(make_batches(rank))
+
+    def forward_backward(batch):
+        x, y = batch
+        loss = ((x @ model.weight.t() - y) ** 2).mean()
+        loss.backward()
+        return loss.detach()
+
+    n_acc = 1 if rank == 0 else 3      # heterogeneous accumulation
+    eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
+                     spec=spec, comm=comm, rank=rank, device=device, opt=opt,
+                     sched=sched, forward_backward=forward_backward,
+                     next_batch=lambda: next(it), n_grad_accumulation=n_acc)
+    opt.init_master_from_buffer(params)
+    eng.trace = []
+    eng.train_acco(TARGET, n_warmup_steps=0)
+    torch.save({"params": params[:D].clone(), "trace": eng.trace,
+                "count": eng.count_grad_tot, "rounds": eng.round_idx},
+               os.path.join(tmpdir, f"res_{rank}.pt"))
+    teardown_worker()
+
+
+def test_acco_heterogeneous_accumulation_ws2():
+    tmpdir = run_distributed(_worker_hetero, 2, timeout=240)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert res[0]["count"] >= TARGET
+    # rank 1 contributed ~3x the grads per round
+    assert sum(res[1]["trace"]) > sum(res[0]["trace"])
+
+    torch.manual_seed(7)
+    w0 = nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone()
+    traces = {r: res[r]["trace"] for r in range(2)}
+    P, count_tot = oracle_replay(w0, traces, 2, 0)
+    assert count_tot == res[0]["count"]
+    assert torch.allclose(P, res[0]["params"], atol=1e-6, rtol=1e-6)
