@@ -8,15 +8,14 @@
 //               dQ += dS·K.
 //   dkv kernel: grid over KV tiles; for each q tile qt>=j accumulate
 //               dV += P^T·dO and dK += dS^T·Q.
-//   Delta[b,h,q] = rowsum(dO ∘ O) is computed by the Python wrapper
-//   (one fused multiply-reduce).
+//   Delta[b,h,q] = rowsum(dO ∘ O) is computed by the Python wrapper.
 //
-// Same fragment-layout playbook as the forward: swapped/non-swapped MFMA
-// picked so every global A/B fragment is a contiguous 16-byte per-lane
-// load; C-layout products that must feed the *A* side of the next MFMA
-// (P, dS) take one per-wave LDS round trip; operands needed transposed on
-// the *B* side (K^T, Q^T, dO^T) are staged transposed in LDS once per tile
-// and shared by the 4 waves.
+// Fragment-layout playbook as in the forward: MFMA operand order chosen so
+// every A/B fragment is a contiguous 16-byte per-lane load; tiles used by
+// all 4 waves are staged ONCE in LDS (row-major straight copies for A
+// fragments; transposed with paired-row b32 writes for B fragments);
+// C-layout products feeding the next MFMA's A side (P, dS) take a
+// per-wave LDS round trip.
 
 #include "common.h"
 
@@ -28,26 +27,41 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr int TILE = 64;
 constexpr int PAD = 8;
-constexpr int LST = TILE + PAD;       // LDS row stride (bf16 elems)
+constexpr int LST = TILE + PAD;       // 72: transposed-tile & P row stride
 
 #define MFMA(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
 
 // stage a [TILE × D] global tile TRANSPOSED into LDS [D][LST]
+// (paired-row ushort2 writes: half the instructions of scalar b16)
 template <int D>
 ACCO_DEV void stage_transposed(const u16* src, long long row_stride,
-                               u16* dst_lds) {
-  const int r = threadIdx.x & 63;                   // tile row
-  for (int dg = threadIdx.x >> 6; dg < D / 16; dg += 4) {
-    const u16* p = src + (long long)r * row_stride + dg * 16;
-    ushort4 a = reinterpret_cast<const ushort4*>(p)[0];
-    ushort4 b = reinterpret_cast<const ushort4*>(p)[1];
-    ushort4 c = reinterpret_cast<const ushort4*>(p)[2];
-    ushort4 d = reinterpret_cast<const ushort4*>(p)[3];
-    u16 vals[16] = {a.x, a.y, a.z, a.w, b.x, b.y, b.z, b.w,
-                    c.x, c.y, c.z, c.w, d.x, d.y, d.z, d.w};
+                               u16* dst) {
+  const int r2 = (threadIdx.x & 31) * 2;            // tile row pair
+  for (int dg = threadIdx.x >> 5; dg < D / 8; dg += 8) {
+    ushort4 a0 = reinterpret_cast<const ushort4*>(
+        src + (long long)r2 * row_stride + dg * 8)[0];
+    ushort4 a1 = reinterpret_cast<const ushort4*>(
+        src + (long long)r2 * row_stride + dg * 8)[1];
+    ushort4 b0 = reinterpret_cast<const ushort4*>(
+        src + (long long)(r2 + 1) * row_stride + dg * 8)[0];
+    ushort4 b1 = reinterpret_cast<const ushort4*>(
+        src + (long long)(r2 + 1) * row_stride + dg * 8)[1];
+    u16 av[8] = {a0.x, a0.y, a0.z, a0.w, a1.x, a1.y, a1.z, a1.w};
+    u16 bv[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
 #pragma unroll
-    for (int i = 0; i < 16; ++i)
-      dst_lds[(dg * 16 + i) * LST + r] = vals[i];
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<ushort2*>(dst + (dg * 8 + i) * LST + r2) =
+          make_ushort2(av[i], bv[i]);
+  }
+}
+
+// stage a [TILE × D] global tile ROW-MAJOR into LDS [TILE][D+8]
+template <int D>
+ACCO_DEV void stage_rowmajor(const u16* src, long long row_stride, u16* dst) {
+  for (int c = threadIdx.x; c < TILE * (D / 8); c += 256) {
+    const int r = c / (D / 8), dc = c % (D / 8);
+    reinterpret_cast<uint4*>(dst + r * (D + 8))[dc] =
+        *reinterpret_cast<const uint4*>(src + (long long)r * row_stride + dc * 8);
   }
 }
 
@@ -62,14 +76,17 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                         int S, int H, int Hkv, float scale, int window) {
   constexpr int KS = D / 32;
   constexpr int DT = D / 16;
+  constexpr int KROW = D + 8;
   const int qt = blockIdx.x, bh = blockIdx.y;
   const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int lg = lane >> 4, lc = lane & 15;
 
   extern __shared__ __attribute__((aligned(16))) u16 smem[];
-  u16* kT_lds = smem;                            // [D][LST] K^T tile
-  u16* ds_lds = smem + D * LST + wave * 16 * LST;  // per-wave dS tile
+  u16* kT_lds = smem;                             // [D][LST]
+  u16* k_row = smem + D * LST;                    // [TILE][KROW]
+  u16* v_row = k_row + TILE * KROW;               // [TILE][KROW]
+  u16* ds_lds = v_row + TILE * KROW + wave * 16 * LST;   // per-wave dS^T
 
   const long long qs = (long long)H * D, ks = (long long)Hkv * D;
   const int q0 = qt * TILE + wave * 16;
@@ -78,15 +95,13 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const u16* Kb = k + (long long)b * S * ks + (long long)hkv * D;
   const u16* Vb = v + (long long)b * S * ks + (long long)hkv * D;
 
-  // per-lane fragments held for the whole kernel:
-  // Q (B of swapped S^T), dO (B of swapped dP^T): lane = [row lc][d lg*8+]
+  // Q / dO as B operands (swapped MFMAs): lane = [row lc][d lg*8+i]
   short8 qf[KS], dof[KS];
 #pragma unroll
   for (int s = 0; s < KS; ++s) {
     qf[s] = *reinterpret_cast<const short8*>(Qp + (long long)lc * qs + s * 32 + lg * 8);
     dof[s] = *reinterpret_cast<const short8*>(dOp + (long long)lc * qs + s * 32 + lg * 8);
   }
-  // per-col (q=lc) stats
   const float lse_c = lse[(long long)bh * S + q0 + lc];
   const float delta_c = delta[(long long)bh * S + q0 + lc];
 
@@ -103,11 +118,11 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   for (int j = j_lo; j <= qt; ++j) {
     __syncthreads();
     stage_transposed<D>(Kb + (long long)(j * TILE) * ks, ks, kT_lds);
+    stage_rowmajor<D>(Kb + (long long)(j * TILE) * ks, ks, k_row);
+    stage_rowmajor<D>(Vb + (long long)(j * TILE) * ks, ks, v_row);
     __syncthreads();
 
-    const u16* Kt = Kb + (long long)(j * TILE) * ks;
-    const u16* Vt = Vb + (long long)(j * TILE) * ks;
-    // S^T = K·Q^T and dP^T = V·dO^T (both C: col=q=lc, row=kv spread)
+    // S^T = K·Q^T and dP^T = V·dO^T (C: col=q=lc, row=kv spread)
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int m16 = 0; m16 < 4; ++m16) {
@@ -115,9 +130,9 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 #pragma unroll
       for (int s = 0; s < KS; ++s) {
         short8 kf = *reinterpret_cast<const short8*>(
-            Kt + (long long)(m16 * 16 + lc) * ks + s * 32 + lg * 8);
+            k_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
         short8 vf = *reinterpret_cast<const short8*>(
-            Vt + (long long)(m16 * 16 + lc) * ks + s * 32 + lg * 8);
+            v_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
         a1 = MFMA(kf, qf[s], a1);
         a2 = MFMA(vf, dof[s], a2);
       }
@@ -151,7 +166,6 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
           ds_lds + lc * LST + s * 32 + lg * 8);
 #pragma unroll
       for (int t = 0; t < DT; ++t) {
-        // B[k=kv][col=d]: lane reads kT_lds[d=t*16+lc][kv=s*32+lg*8 ..]
         short8 kb = *reinterpret_cast<const short8*>(
             kT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
         acc_dq[t] = MFMA(dsa, kb, acc_dq[t]);
@@ -177,12 +191,10 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                          const float* __restrict__ lse,
                          const float* __restrict__ delta,
                          u16* __restrict__ dk, u16* __restrict__ dv,
-                         int S, int H, int Hkv, float scale, int window,
-                         int kv_repeat) {
-  // one workgroup per (kv tile, b, h-of-Q); grads for the SHARED kv head
-  // are accumulated over its kv_repeat query heads by h-loop inside.
+                         int S, int H, int Hkv, float scale, int window) {
   constexpr int KS = D / 32;
   constexpr int DT = D / 16;
+  constexpr int KROW = D + 8;
   const int j = blockIdx.x, bh = blockIdx.y;
   const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -191,8 +203,10 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   extern __shared__ __attribute__((aligned(16))) u16 smem[];
   u16* qT_lds = smem;                               // [D][LST]
   u16* doT_lds = smem + D * LST;                    // [D][LST]
-  u16* p_lds = smem + 2 * D * LST + wave * 16 * LST;      // per-wave P^T
-  u16* ds_lds = smem + 2 * D * LST + (4 + wave) * 16 * LST;  // per-wave dS^T
+  u16* q_row = doT_lds + D * LST;                   // [TILE][KROW]
+  u16* do_row = q_row + TILE * KROW;                // [TILE][KROW]
+  u16* p_lds = do_row + TILE * KROW + wave * 16 * LST;
+  u16* ds_lds = do_row + TILE * KROW + (4 + wave) * 16 * LST;
 
   const long long qs = (long long)H * D, ks = (long long)Hkv * D;
   const int kv0 = j * TILE + wave * 16;             // wave's 16 kv rows
@@ -201,8 +215,7 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const u16* Qb = q + (long long)b * S * qs + (long long)h * D;
   const u16* dOb = dO + (long long)b * S * qs + (long long)h * D;
 
-  // K, V fragments as B operands (transposed reads are contiguous):
-  // B[k=d][col=kv]: lane = K[kv=lc][d=lg*8+i]
+  // K, V as B operands: lane = [row kv=lc][d lg*8+i]
   short8 kTf[KS], vTf[KS];
 #pragma unroll
   for (int s = 0; s < KS; ++s) {
@@ -219,8 +232,7 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 
   int qt_hi = S / TILE - 1;
   if (window > 0) {
-    // largest q tile with q - window < kv_max  →  q < kv_max + window
-    const int q_lim = j * TILE + TILE - 1 + window;       // exclusive-ish
+    const int q_lim = j * TILE + TILE - 1 + window;
     qt_hi = min(qt_hi, q_lim / TILE);
   }
 
@@ -228,12 +240,11 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     __syncthreads();
     stage_transposed<D>(Qb + (long long)(qt * TILE) * qs, qs, qT_lds);
     stage_transposed<D>(dOb + (long long)(qt * TILE) * qs, qs, doT_lds);
+    stage_rowmajor<D>(Qb + (long long)(qt * TILE) * qs, qs, q_row);
+    stage_rowmajor<D>(dOb + (long long)(qt * TILE) * qs, qs, do_row);
     __syncthreads();
 
-    const u16* Qt = Qb + (long long)(qt * TILE) * qs;
-    const u16* dOt = dOb + (long long)(qt * TILE) * qs;
-
-    // S = Q·K^T, dP = dO·V^T (C: col=kv=lc, row=q spread)
+    // S = Q·K^T, dP = dO·V^T (C: col = kv = lc, row = q spread)
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int m16 = 0; m16 < 4; ++m16) {
@@ -241,21 +252,15 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 #pragma unroll
       for (int s = 0; s < KS; ++s) {
         short8 qfr = *reinterpret_cast<const short8*>(
-            Qt + (long long)(m16 * 16 + lc) * qs + s * 32 + lg * 8);
+            q_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
         short8 dofr = *reinterpret_cast<const short8*>(
-            dOt + (long long)(m16 * 16 + lc) * qs + s * 32 + lg * 8);
-        a1 = MFMA(qfr, kTf[s], a1);    // wait: A rows must be q…
+            do_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+        a1 = MFMA(qfr, kTf[s], a1);
         a2 = MFMA(dofr, vTf[s], a2);
       }
       st[m16] = a1;
       dpt[m16] = a2;
     }
-    // NOTE on layout: A = Q fragment has rows q = m16*16+lc?  No:
-    // A[row][k] with row=lane&15 — the m16 loop walks q sub-tiles, so C of
-    // sub-tile m16 covers q rows m16*16..+15 with col=kv=lc … but C's col
-    // belongs to B (K^T cols = the wave's 16 kv rows).  See python-side
-    // layout test; the equations below use: C[m16][r]: q_g = qt*TILE +
-    // m16*16 + lg*4 + r, kv_g = kv0 + lc.
 
     const int kv_g = kv0 + lc;
 #pragma unroll
@@ -281,8 +286,8 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
     __builtin_amdgcn_s_waitcnt(0);
 
-    // dV += P^T·dO : A = P^T[kv=lc][q], B = dO^T[q][d] from doT_lds
-    // dK += dS^T·Q : A = dS^T[kv=lc][q], B = Q^T[q][d] from qT_lds
+    // dV += P^T·dO (A: P^T[kv=lc][q], B: dO^T from doT_lds)
+    // dK += dS^T·Q (A: dS^T[kv=lc][q], B: Q^T from qT_lds)
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       short8 pa = *reinterpret_cast<const short8*>(
@@ -302,9 +307,7 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   }
 
   // store dK/dV rows kv = kv0 + lg*4 + r, col d = t*16+lc.
-  // GQA (kv_repeat > 1): several query heads share this kv head → the
-  // wrapper allocates per-QUERY-head buffers [B,S,H,D] and reduces over
-  // the group afterwards, so each workgroup writes its own h slice.
+  // GQA: emitted per QUERY head; the wrapper group-reduces to kv heads.
   u16* dKp = dk + ((long long)b * S + kv0) * qs + (long long)h * D;
   u16* dVp = dv + ((long long)b * S + kv0) * qs + (long long)h * D;
 #pragma unroll
@@ -316,7 +319,6 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
       dVp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
           f32_to_bf16(acc_dv[t][r]);
     }
-  (void)kv_repeat;
 }
 
 }  // namespace
@@ -328,7 +330,7 @@ void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       void* dq, int B, int S, int H, int Hkv, int D,
                       float scale, int window, hipStream_t stream) {
   dim3 grid(S / TILE, B * H);
-  const int lds = (D + 4 * 16) * LST * sizeof(u16);
+  const int lds = (D * LST + 2 * TILE * (D + 8) + 4 * 16 * LST) * sizeof(u16);
   if (D == 64)
     hipLaunchKernelGGL(attn_bwd_dq_kernel<64>, grid, dim3(256), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v,
@@ -346,18 +348,18 @@ void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        void* dk, void* dv, int B, int S, int H, int Hkv,
                        int D, float scale, int window, hipStream_t stream) {
   dim3 grid(S / TILE, B * H);
-  const int lds = (2 * D + 2 * 4 * 16) * LST * sizeof(u16);
-  const int rep = H / Hkv;
+  const int lds =
+      (2 * D * LST + 2 * TILE * (D + 8) + 8 * 16 * LST) * sizeof(u16);
   if (D == 64)
     hipLaunchKernelGGL(attn_bwd_dkv_kernel<64>, grid, dim3(256), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v,
                        (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                       Hkv, scale, window, rep);
+                       Hkv, scale, window);
   else
     hipLaunchKernelGGL(attn_bwd_dkv_kernel<128>, grid, dim3(256), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v,
                        (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                       Hkv, scale, window, rep);
+                       Hkv, scale, window);
 }
 
 }  // extern "C"

@@ -55,10 +55,12 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const int lg = lane >> 4;      // lane group 0..3
   const int lc = lane & 15;      // col / row-in-16 index
 
-  // LDS: V^T tile [D][72] shared; P tiles [16][72] per wave
+  // LDS: V^T tile [D][72] + K tile [64][D+8] shared; P tiles per wave
   extern __shared__ __attribute__((aligned(16))) u16 smem[];
   u16* v_lds = smem;                                   // D * LSTRIDE
-  u16* p_lds = smem + D * LSTRIDE + wave * 16 * LSTRIDE;
+  u16* k_lds = smem + D * LSTRIDE;                     // KT * (D + 8)
+  u16* p_lds = k_lds + KT * (D + 8) + wave * 16 * LSTRIDE;
+  constexpr int KROW = D + 8;                          // K tile row stride
 
   const int q0 = qt * QT + wave * 16;                  // wave's first q row
   const long long qrow_stride = (long long)H * D;
@@ -89,44 +91,53 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const int j_hi = qt;           // causal
 
   for (int j = j_lo; j <= j_hi; ++j) {
-    // ---- stage V^T tile cooperatively (4 waves share)
+    // ---- stage V^T (transposed, b32 paired-kv writes) and K (row-major
+    // straight copy) cooperatively — shared by the 4 waves, so each K/V
+    // element crosses HBM/L2 once instead of once per wave
     __syncthreads();             // previous tile's reads done
     {
       const u16* Vt = Vb + (long long)(j * KT) * krow_stride;
-      // 256 threads: thread handles (kv = tid%64, dgroup = tid/64)
-      const int kv = threadIdx.x & 63;
-      for (int dg = threadIdx.x >> 6; dg < D / 16; dg += 4) {
-        const u16* src = Vt + (long long)kv * krow_stride + dg * 16;
-        ushort4 a = reinterpret_cast<const ushort4*>(src)[0];
-        ushort4 c = reinterpret_cast<const ushort4*>(src)[1];
-        u16 vals[8] = {a.x, a.y, a.z, a.w, c.x, c.y, c.z, c.w};
-        ushort4 a2 = reinterpret_cast<const ushort4*>(src + 8)[0];
-        ushort4 c2 = reinterpret_cast<const ushort4*>(src + 8)[1];
-        u16 vals2[8] = {a2.x, a2.y, a2.z, a2.w, c2.x, c2.y, c2.z, c2.w};
+      // V: thread handles kv pair (2*(tid%32), +1), d-group of 8 = tid/32
+      const int kv2 = (threadIdx.x & 31) * 2;
+      for (int dg = threadIdx.x >> 5; dg < D / 8; dg += 8) {
+        ushort4 a0 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)kv2 * krow_stride + dg * 8)[0];
+        ushort4 a1 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)kv2 * krow_stride + dg * 8)[1];
+        ushort4 b0 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)(kv2 + 1) * krow_stride + dg * 8)[0];
+        ushort4 b1 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)(kv2 + 1) * krow_stride + dg * 8)[1];
+        u16 av[8] = {a0.x, a0.y, a0.z, a0.w, a1.x, a1.y, a1.z, a1.w};
+        u16 bv[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          v_lds[(dg * 16 + i) * LSTRIDE + kv] = vals[i];
-          v_lds[(dg * 16 + 8 + i) * LSTRIDE + kv] = vals2[i];
-        }
+        for (int i = 0; i < 8; ++i)
+          *reinterpret_cast<ushort2*>(v_lds + (dg * 8 + i) * LSTRIDE + kv2) =
+              make_ushort2(av[i], bv[i]);
+      }
+      // K: straight vector copy into [64][KROW]
+      const u16* Kt = Kb + (long long)(j * KT) * krow_stride;
+      for (int c = threadIdx.x; c < KT * (D / 8); c += 256) {
+        const int kv = c / (D / 8), dc = c % (D / 8);
+        reinterpret_cast<uint4*>(k_lds + kv * KROW)[dc] =
+            *reinterpret_cast<const uint4*>(
+                Kt + (long long)kv * krow_stride + dc * 8);
       }
     }
     __syncthreads();
 
     // ---- S^T tile: st[m16] = K_sub · Q^T  (C: col=q=lc, row=kv spread)
     f32x4 st[4];
-    {
-      const u16* Kt = Kb + (long long)(j * KT) * krow_stride;
 #pragma unroll
-      for (int m16 = 0; m16 < 4; ++m16) {
-        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int m16 = 0; m16 < 4; ++m16) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int s = 0; s < KS; ++s) {
-          short8 kf = *reinterpret_cast<const short8*>(
-              Kt + (long long)(m16 * 16 + lc) * krow_stride + s * 32 + lg * 8);
-          acc = MFMA(kf, qf[s], acc);
-        }
-        st[m16] = acc;
+      for (int s = 0; s < KS; ++s) {
+        short8 kf = *reinterpret_cast<const short8*>(
+            k_lds + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+        acc = MFMA(kf, qf[s], acc);
       }
+      st[m16] = acc;
     }
 
     // ---- mask + online softmax (per col q=lc)
@@ -229,7 +240,8 @@ extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               int Hkv, int D, float scale, int window,
                               hipStream_t stream) {
   dim3 grid(S / QT, B * H);
-  const int lds_bytes = (D + 4 * 16) * LSTRIDE * sizeof(u16);
+  const int lds_bytes =
+      (D * LSTRIDE + KT * (D + 8) + 4 * 16 * LSTRIDE) * sizeof(u16);
   if (D == 64)
     hipLaunchKernelGGL(attn_fwd_kernel<64>, grid, dim3(256), lds_bytes,
                        stream, (const u16*)q, (const u16*)k, (const u16*)v,
