@@ -76,6 +76,10 @@ class DecoupledTrainer:
         # C1: average random-init weights across ranks (reference :180)
         self.comm.all_reduce_avg(self.params)
 
+        # fused QKV / gate-up GEMMs over arena-adjacent weight views
+        from acco_amd.models.fuse import install_fused_projections
+        install_fused_projections(model, self.params, self.grads)
+
         # ---- observability
         out_dir = os.getcwd()
         self.scalars = ScalarLogger(os.path.join(out_dir, "scalars"),

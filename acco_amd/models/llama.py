@@ -41,15 +41,25 @@ class LlamaAttention(nn.Module):
         self.k_proj = nn.Linear(d, cfg.num_kv_heads * hd, bias=False)
         self.v_proj = nn.Linear(d, cfg.num_kv_heads * hd, bias=False)
         self.o_proj = nn.Linear(cfg.num_heads * hd, d, bias=False)
+        self._fused_qkv = None    # (w_view, g_view, splits) via models.fuse
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
         cfg = self.cfg
         # keep everything in the projections' natural [B, S, H, D] layout:
         # RoPE and attention kernels walk D coalesced, no transpose copies
-        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
-        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        if self._fused_qkv is not None:
+            from acco_amd.models.fuse import FusedArenaLinearFn
+            w, g, splits = self._fused_qkv
+            qkv = FusedArenaLinearFn.apply(x, w, g)
+            q, k, v = torch.split(qkv, splits, dim=-1)
+            q = q.contiguous().view(B, S, cfg.num_heads, cfg.head_dim)
+            k = k.contiguous().view(B, S, cfg.num_kv_heads, cfg.head_dim)
+            v = v.contiguous().view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        else:
+            q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
+            k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+            v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
         q, k = ops.rope_apply(q, k, cos, sin)
         o = ops.causal_attention(q, k, v)          # [B, S, H, D]
         return self.o_proj(o.reshape(B, S, -1))
@@ -61,8 +71,16 @@ class LlamaMLP(nn.Module):
         self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
         self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
         self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+        self._fused_gate_up = None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self._fused_gate_up is not None:
+            from acco_amd.models.fuse import FusedArenaLinearFn
+            w, g, splits = self._fused_gate_up
+            gu = FusedArenaLinearFn.apply(x, w, g)
+            gate, up = torch.split(gu, splits, dim=-1)
+            return self.down_proj(ops.swiglu(gate.contiguous(),
+                                             up.contiguous()))
         return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
 
 

@@ -23,6 +23,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -114,6 +115,11 @@ def main():
     grads = arena.attach_grad_arena(model, dtype, device, pad_to=spec.total)
     comm = CommBackend(device)
     comm.all_reduce_avg(params)
+
+    from acco_amd.models.fuse import install_fused_projections
+    n_fused = install_fused_projections(model, params, grads)
+    if rank == 0:
+        print(f"# fused projection groups: {n_fused}", file=sys.stderr)
 
     opt = ShardedAdamW(spec, rank, device, lr=args.lr, betas=(0.9, 0.95),
                        eps=1e-8, weight_decay=0.1)
