@@ -103,6 +103,16 @@ class AccoEngine:
         # counter for token accounting
         self.on_round_boundary: Optional[Callable[[int], None]] = None
         self.micro_steps = 0
+        # com-log: per-round wall time of the communication round, dumped by
+        # the trainer at end of training (the reference's save_com_logs
+        # channel, utils/logs_utils.py:141-152)
+        self.com_log: list = []
+        # ACCO_DEBUG_HANDOFF=1: checksum the com-buffer ownership handoff
+        # between the comm and compute threads (SURVEY.md §5 race-detection
+        # rebuild note)
+        import os as _os
+        self._debug_handoff = _os.environ.get("ACCO_DEBUG_HANDOFF") == "1"
+        self._handoff_sum: Optional[torch.Tensor] = None
 
     # ------------------------------------------------------------ pieces
 
@@ -120,6 +130,11 @@ class AccoEngine:
         """Swap: params ← buffer (new params); buffer ← accumulated grads;
         publish local grad count; optionally zero grads+count
         (reference update_buffers_step :43-63)."""
+        if self._debug_handoff and self._handoff_sum is not None:
+            now = self.com_buffer[:min(self.n, 1 << 20)].float().sum()
+            assert torch.equal(now, self._handoff_sum), (
+                "com-buffer ownership handoff violated: buffer changed "
+                "between com_finished and update_buffers_step")
         self.params[:self.n].copy_(self.com_buffer[:self.n])
         self.com_buffer[:self.n].copy_(self.grads[:self.n])
         if self.com_buffer.numel() > self.n:
@@ -140,8 +155,14 @@ class AccoEngine:
         count of the round."""
         if advance_sched is None:
             advance_sched = commit
+        t0 = time.time()
         with trace_range("acco/communication_round"):
-            return self._communication_round(commit, advance_sched)
+            n = self._communication_round(commit, advance_sched)
+        self.com_log.append({"round": self.round_idx, "commit": commit,
+                             "t": time.time() - t0, "count": n})
+        if self._debug_handoff:
+            self._handoff_sum = self.com_buffer[:min(self.n, 1 << 20)].float().sum()
+        return n
 
     @torch.no_grad()
     def _communication_round(self, commit: bool, advance_sched: bool) -> int:

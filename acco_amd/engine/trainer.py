@@ -355,6 +355,12 @@ class DecoupledTrainer:
 
     def _finalize(self) -> None:
         total_time = time.time() - self.t_beg
+        if self.rank == 0 and self.engine is not None and self.engine.com_log:
+            # com-log dump (reference save_com_logs, utils/logs_utils.py:141-152)
+            import json
+            with open(os.path.join(os.getcwd(),
+                                   f"com_logs_{self.id_run}.json"), "w") as f:
+                json.dump(self.engine.com_log, f)
         if self.rank == 0:
             loss = (float(self.engine.loss_static.item())
                     if self.engine else 0.0)
