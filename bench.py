@@ -46,6 +46,13 @@ MODELS = {
                  max_position_embeddings=2048, tie_word_embeddings=True),
 }
 
+# GPT-Neo family entries (BASELINE.json configs 1-2: GPT-2-small-shaped)
+NEO_MODELS = {
+    "gptneo-125m": dict(hidden_size=768, num_layers=12, num_heads=12,
+                        vocab_size=50304, max_position_embeddings=2048,
+                        window_size=256),
+}
+
 
 def parse_args():
     p = argparse.ArgumentParser()
@@ -98,16 +105,20 @@ def main():
     from acco_amd.engine.acco import AccoEngine
     from acco_amd.engine.scheduler import LRSchedule
     from acco_amd.engine.sharded_adamw import ShardedAdamW
-    from acco_amd.models import LlamaConfig, LlamaForCausalLM
+    from acco_amd.models import (GPTNeoConfig, GPTNeoForCausalLM,
+                                 LlamaConfig, LlamaForCausalLM)
     from acco_amd.parallel.comm import CommBackend, ShardSpec
     from acco_amd.parallel.ddp import NativeZeroDDP
 
     model_name = args.model if cuda else "tiny"
-    mcfg = LlamaConfig(**MODELS[model_name])
-    dtype = torch.bfloat16 if cuda else torch.float32
-
     torch.manual_seed(42)
-    model = LlamaForCausalLM(mcfg)
+    if model_name in NEO_MODELS:
+        mcfg = GPTNeoConfig(**NEO_MODELS[model_name])
+        model = GPTNeoForCausalLM(mcfg)
+    else:
+        mcfg = LlamaConfig(**MODELS[model_name])
+        model = LlamaForCausalLM(mcfg)
+    dtype = torch.bfloat16 if cuda else torch.float32
 
     n_live = arena.live_numel(model)
     spec = ShardSpec.build(n_live, world, buckets=args.buckets)
