@@ -67,7 +67,7 @@ ACCO_DEV void stage_rowmajor(const u16* src, long long row_stride, u16* dst) {
 }
 
 // ------------------------------------------------------------------- dQ
-template <int D>
+template <int D, int QW>
 __global__ __launch_bounds__(256)
 void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                         const u16* __restrict__ v, const u16* __restrict__ dO,
@@ -78,6 +78,8 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   constexpr int KS = D / 32;
   constexpr int DT = D / 16;
   constexpr int KROW = D + 8;
+  constexpr int M2 = QW / 16;        // q sub-tiles per wave
+  constexpr int QT = 4 * QW;         // q rows per workgroup
   const int qt = blockIdx.x, bh = blockIdx.y;
   const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -90,102 +92,117 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   u16* ds_lds = v_row + TILE * KROW + wave * 16 * LST;   // per-wave dS^T
 
   const long long qs = (long long)H * D, ks = (long long)Hkv * D;
-  const int q0 = qt * TILE + wave * 16;
+  const int q0 = qt * QT + wave * QW;
   const u16* Qp = q + ((long long)b * S + q0) * qs + (long long)h * D;
   const u16* dOp = dO + ((long long)b * S + q0) * qs + (long long)h * D;
   const u16* Kb = k + (long long)b * S * ks + (long long)hkv * D;
   const u16* Vb = v + (long long)b * S * ks + (long long)hkv * D;
 
-  // Q / dO as B operands (swapped MFMAs): lane = [row lc][d lg*8+i]
-  short8 qf[KS], dof[KS];
+  // Q / dO as B operands (swapped MFMAs): lane = [row m*16+lc][d lg*8+i]
+  short8 qf[M2][KS], dof[M2][KS];
+  float lse_c[M2], delta_c[M2];
 #pragma unroll
-  for (int s = 0; s < KS; ++s) {
-    qf[s] = *reinterpret_cast<const short8*>(Qp + (long long)lc * qs + s * 32 + lg * 8);
-    dof[s] = *reinterpret_cast<const short8*>(dOp + (long long)lc * qs + s * 32 + lg * 8);
+  for (int m = 0; m < M2; ++m) {
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      qf[m][s] = *reinterpret_cast<const short8*>(
+          Qp + (long long)(m * 16 + lc) * qs + s * 32 + lg * 8);
+      dof[m][s] = *reinterpret_cast<const short8*>(
+          dOp + (long long)(m * 16 + lc) * qs + s * 32 + lg * 8);
+    }
+    lse_c[m] = lse[(long long)bh * S + q0 + m * 16 + lc];
+    delta_c[m] = delta[(long long)bh * S + q0 + m * 16 + lc];
   }
-  const float lse_c = lse[(long long)bh * S + q0 + lc];
-  const float delta_c = delta[(long long)bh * S + q0 + lc];
 
-  f32x4 acc_dq[DT];
+  f32x4 acc_dq[M2][DT];
 #pragma unroll
-  for (int t = 0; t < DT; ++t) acc_dq[t] = {0.f, 0.f, 0.f, 0.f};
+  for (int m = 0; m < M2; ++m)
+#pragma unroll
+    for (int t = 0; t < DT; ++t) acc_dq[m][t] = {0.f, 0.f, 0.f, 0.f};
 
   int j_lo = 0;
   if (window > 0) {
-    int kv_min = qt * TILE - window + 1;
+    int kv_min = qt * QT - window + 1;
     if (kv_min > 0) j_lo = kv_min / TILE;
   }
+  const int j_hi = (qt * QT + QT - 1) / TILE;
 
-  for (int j = j_lo; j <= qt; ++j) {
+  for (int j = j_lo; j <= j_hi; ++j) {
     __syncthreads();
     stage_transposed<D>(Kb + (long long)(j * TILE) * ks, ks, kT_lds);
     stage_rowmajor<D>(Kb + (long long)(j * TILE) * ks, ks, k_row);
     stage_rowmajor<D>(Vb + (long long)(j * TILE) * ks, ks, v_row);
     __syncthreads();
 
-    // S^T = K·Q^T and dP^T = V·dO^T (C: col=q=lc, row=kv spread)
-    f32x4 st[4], dpt[4];
 #pragma unroll
-    for (int m16 = 0; m16 < 4; ++m16) {
-      f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+    for (int m = 0; m < M2; ++m) {
+      // S^T = K·Q^T and dP^T = V·dO^T (C: col=q=lc, row=kv spread)
+      f32x4 st[4], dpt[4];
 #pragma unroll
-      for (int s = 0; s < KS; ++s) {
-        short8 kf = *reinterpret_cast<const short8*>(
-            k_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
-        short8 vf = *reinterpret_cast<const short8*>(
-            v_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
-        a1 = MFMA(kf, qf[s], a1);
-        a2 = MFMA(vf, dof[s], a2);
+      for (int m16 = 0; m16 < 4; ++m16) {
+        f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          short8 kf = *reinterpret_cast<const short8*>(
+              k_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+          short8 vf = *reinterpret_cast<const short8*>(
+              v_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+          a1 = MFMA(kf, qf[m][s], a1);
+          a2 = MFMA(vf, dof[m][s], a2);
+        }
+        st[m16] = a1;
+        dpt[m16] = a2;
       }
-      st[m16] = a1;
-      dpt[m16] = a2;
-    }
 
-    // dS^T = P^T ∘ (dP^T - Delta) * scale, P = exp(S*scale - lse)
-    const int q_g = qt * TILE + wave * 16 + lc;
+      // dS^T = P^T ∘ (dP^T - Delta) * scale, P = exp(S*scale - lse)
+      const int q_g = q0 + m * 16 + lc;
 #pragma unroll
-    for (int m16 = 0; m16 < 4; ++m16) {
-      u16 pk[4];
+      for (int m16 = 0; m16 < 4; ++m16) {
+        u16 pk[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int kv_g = j * TILE + m16 * 16 + lg * 4 + r;
-        bool valid = (kv_g <= q_g);
-        if (window > 0) valid = valid && (kv_g > q_g - window);
-        const float pval = valid ? __expf(st[m16][r] * scale - lse_c) : 0.0f;
-        const float dsv = pval * (dpt[m16][r] - delta_c) * scale;
-        pk[r] = f32_to_bf16(dsv);
+        for (int r = 0; r < 4; ++r) {
+          const int kv_g = j * TILE + m16 * 16 + lg * 4 + r;
+          bool valid = (kv_g <= q_g);
+          if (window > 0) valid = valid && (kv_g > q_g - window);
+          const float pval =
+              valid ? __expf(st[m16][r] * scale - lse_c[m]) : 0.0f;
+          const float dsv = pval * (dpt[m16][r] - delta_c[m]) * scale;
+          pk[r] = f32_to_bf16(dsv);
+        }
+        *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
+            make_ushort4(pk[0], pk[1], pk[2], pk[3]);
       }
-      *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
-          make_ushort4(pk[0], pk[1], pk[2], pk[3]);
-    }
-    __builtin_amdgcn_s_waitcnt(0);   // lgkm: own-wave LDS writes
+      __builtin_amdgcn_s_waitcnt(0);   // lgkm: own-wave LDS writes
 
-    // dQ += dS·K : A = dS[q=lc][kv], B = K^T[kv][d] from kT_lds
+      // dQ += dS·K : A = dS[q=lc][kv], B = K^T[kv][d] from kT_lds
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      short8 dsa = *reinterpret_cast<const short8*>(
-          ds_lds + lc * LST + s * 32 + lg * 8);
+      for (int s = 0; s < 2; ++s) {
+        short8 dsa = *reinterpret_cast<const short8*>(
+            ds_lds + lc * LST + s * 32 + lg * 8);
 #pragma unroll
-      for (int t = 0; t < DT; ++t) {
-        short8 kb = *reinterpret_cast<const short8*>(
-            kT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
-        acc_dq[t] = MFMA(dsa, kb, acc_dq[t]);
+        for (int t = 0; t < DT; ++t) {
+          short8 kb = *reinterpret_cast<const short8*>(
+              kT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+          acc_dq[m][t] = MFMA(dsa, kb, acc_dq[m][t]);
+        }
       }
     }
   }
 
-  // store dQ rows q=lg*4+r, col d=t*16+lc
+  // store dQ rows q = m*16 + lg*4+r, col d=t*16+lc
   u16* dQp = dq + ((long long)b * S + q0) * qs + (long long)h * D;
 #pragma unroll
-  for (int r = 0; r < 4; ++r)
+  for (int m = 0; m < M2; ++m)
 #pragma unroll
-    for (int t = 0; t < DT; ++t)
-      dQp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
-          f32_to_bf16(acc_dq[t][r]);
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int t = 0; t < DT; ++t)
+        dQp[(long long)(m * 16 + lg * 4 + r) * qs + t * 16 + lc] =
+            f32_to_bf16(acc_dq[m][t][r]);
 }
 
 // ---------------------------------------------------------------- dK, dV
-template <int D>
+template <int D, int QW>
 __global__ __launch_bounds__(256)
 void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                          const u16* __restrict__ v, const u16* __restrict__ dO,
@@ -196,7 +213,9 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   constexpr int KS = D / 32;
   constexpr int DT = D / 16;
   constexpr int KROW = D + 8;
-  const int j = blockIdx.x, bh = blockIdx.y;
+  constexpr int M2 = QW / 16;       // kv sub-tiles per wave
+  constexpr int KB = 4 * QW;        // kv rows per workgroup
+  const int jb = blockIdx.x, bh = blockIdx.y;
   const int b = bh / H, h = bh % H, hkv = h / (H / Hkv);
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int lg = lane >> 4, lc = lane & 15;
@@ -210,34 +229,41 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   u16* ds_lds = do_row + TILE * KROW + (4 + wave) * 16 * LST;
 
   const long long qs = (long long)H * D, ks = (long long)Hkv * D;
-  const int kv0 = j * TILE + wave * 16;             // wave's 16 kv rows
+  const int kv0 = jb * KB + wave * QW;              // wave's QW kv rows
   const u16* Kp = k + ((long long)b * S + kv0) * ks + (long long)hkv * D;
   const u16* Vp = v + ((long long)b * S + kv0) * ks + (long long)hkv * D;
   const u16* Qb = q + (long long)b * S * qs + (long long)h * D;
   const u16* dOb = dO + (long long)b * S * qs + (long long)h * D;
 
-  // K, V as B operands: lane = [row kv=lc][d lg*8+i]
-  short8 kTf[KS], vTf[KS];
+  // K, V as B operands: lane = [row kv = m*16+lc][d lg*8+i]
+  short8 kTf[M2][KS], vTf[M2][KS];
 #pragma unroll
-  for (int s = 0; s < KS; ++s) {
-    kTf[s] = *reinterpret_cast<const short8*>(Kp + (long long)lc * ks + s * 32 + lg * 8);
-    vTf[s] = *reinterpret_cast<const short8*>(Vp + (long long)lc * ks + s * 32 + lg * 8);
-  }
+  for (int m = 0; m < M2; ++m)
+#pragma unroll
+    for (int s = 0; s < KS; ++s) {
+      kTf[m][s] = *reinterpret_cast<const short8*>(
+          Kp + (long long)(m * 16 + lc) * ks + s * 32 + lg * 8);
+      vTf[m][s] = *reinterpret_cast<const short8*>(
+          Vp + (long long)(m * 16 + lc) * ks + s * 32 + lg * 8);
+    }
 
-  f32x4 acc_dk[DT], acc_dv[DT];
+  f32x4 acc_dk[M2][DT], acc_dv[M2][DT];
 #pragma unroll
-  for (int t = 0; t < DT; ++t) {
-    acc_dk[t] = {0.f, 0.f, 0.f, 0.f};
-    acc_dv[t] = {0.f, 0.f, 0.f, 0.f};
-  }
+  for (int m = 0; m < M2; ++m)
+#pragma unroll
+    for (int t = 0; t < DT; ++t) {
+      acc_dk[m][t] = {0.f, 0.f, 0.f, 0.f};
+      acc_dv[m][t] = {0.f, 0.f, 0.f, 0.f};
+    }
 
   int qt_hi = S / TILE - 1;
   if (window > 0) {
-    const int q_lim = j * TILE + TILE - 1 + window;
+    const int q_lim = jb * KB + KB - 1 + window;
     qt_hi = min(qt_hi, q_lim / TILE);
   }
+  const int qt_lo = (jb * KB) / TILE;
 
-  for (int qt = j; qt <= qt_hi; ++qt) {
+  for (int qt = qt_lo; qt <= qt_hi; ++qt) {
     __syncthreads();
     stage_transposed<D>(Qb + (long long)(qt * TILE) * qs, qs, qT_lds);
     stage_transposed<D>(dOb + (long long)(qt * TILE) * qs, qs, doT_lds);
@@ -245,81 +271,86 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     stage_rowmajor<D>(dOb + (long long)(qt * TILE) * qs, qs, do_row);
     __syncthreads();
 
-    // S = Q·K^T, dP = dO·V^T (C: col = kv = lc, row = q spread)
-    f32x4 st[4], dpt[4];
 #pragma unroll
-    for (int m16 = 0; m16 < 4; ++m16) {
-      f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+    for (int m = 0; m < M2; ++m) {
+      // S = Q·K^T, dP = dO·V^T (C: col = kv = lc, row = q spread)
+      f32x4 st[4], dpt[4];
 #pragma unroll
-      for (int s = 0; s < KS; ++s) {
-        short8 qfr = *reinterpret_cast<const short8*>(
-            q_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
-        short8 dofr = *reinterpret_cast<const short8*>(
-            do_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
-        a1 = MFMA(qfr, kTf[s], a1);
-        a2 = MFMA(dofr, vTf[s], a2);
+      for (int m16 = 0; m16 < 4; ++m16) {
+        f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          short8 qfr = *reinterpret_cast<const short8*>(
+              q_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+          short8 dofr = *reinterpret_cast<const short8*>(
+              do_row + (m16 * 16 + lc) * KROW + s * 32 + lg * 8);
+          a1 = MFMA(qfr, kTf[m][s], a1);
+          a2 = MFMA(dofr, vTf[m][s], a2);
+        }
+        st[m16] = a1;
+        dpt[m16] = a2;
       }
-      st[m16] = a1;
-      dpt[m16] = a2;
-    }
 
-    const int kv_g = kv0 + lc;
+      const int kv_g = kv0 + m * 16 + lc;
 #pragma unroll
-    for (int m16 = 0; m16 < 4; ++m16) {
-      u16 ppk[4], dsk[4];
+      for (int m16 = 0; m16 < 4; ++m16) {
+        u16 ppk[4], dsk[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q_g = qt * TILE + m16 * 16 + lg * 4 + r;
-        bool valid = (kv_g <= q_g) && (q_g < S);
-        if (window > 0) valid = valid && (kv_g > q_g - window);
-        const float lse_q = lse[(long long)bh * S + min(q_g, S - 1)];
-        const float del_q = delta[(long long)bh * S + min(q_g, S - 1)];
-        const float pval = valid ? __expf(st[m16][r] * scale - lse_q) : 0.0f;
-        const float dsv = pval * (dpt[m16][r] - del_q) * scale;
-        ppk[r] = f32_to_bf16(pval);
-        dsk[r] = f32_to_bf16(dsv);
+        for (int r = 0; r < 4; ++r) {
+          const int q_g = qt * TILE + m16 * 16 + lg * 4 + r;
+          bool valid = (kv_g <= q_g) && (q_g < S);
+          if (window > 0) valid = valid && (kv_g > q_g - window);
+          const float lse_q = lse[(long long)bh * S + min(q_g, S - 1)];
+          const float del_q = delta[(long long)bh * S + min(q_g, S - 1)];
+          const float pval = valid ? __expf(st[m16][r] * scale - lse_q) : 0.0f;
+          const float dsv = pval * (dpt[m16][r] - del_q) * scale;
+          ppk[r] = f32_to_bf16(pval);
+          dsk[r] = f32_to_bf16(dsv);
+        }
+        // store P^T and dS^T: row kv=lc, col q = m16*16 + lg*4 + r
+        *reinterpret_cast<ushort4*>(p_lds + lc * LST + m16 * 16 + lg * 4) =
+            make_ushort4(ppk[0], ppk[1], ppk[2], ppk[3]);
+        *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
+            make_ushort4(dsk[0], dsk[1], dsk[2], dsk[3]);
       }
-      // store P^T and dS^T: row kv=lc, col q = m16*16 + lg*4 + r
-      *reinterpret_cast<ushort4*>(p_lds + lc * LST + m16 * 16 + lg * 4) =
-          make_ushort4(ppk[0], ppk[1], ppk[2], ppk[3]);
-      *reinterpret_cast<ushort4*>(ds_lds + lc * LST + m16 * 16 + lg * 4) =
-          make_ushort4(dsk[0], dsk[1], dsk[2], dsk[3]);
-    }
-    __builtin_amdgcn_s_waitcnt(0);
+      __builtin_amdgcn_s_waitcnt(0);
 
-    // dV += P^T·dO (A: P^T[kv=lc][q], B: dO^T from doT_lds)
-    // dK += dS^T·Q (A: dS^T[kv=lc][q], B: Q^T from qT_lds)
+      // dV += P^T·dO (A: P^T[kv=lc][q], B: dO^T from doT_lds)
+      // dK += dS^T·Q (A: dS^T[kv=lc][q], B: Q^T from qT_lds)
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      short8 pa = *reinterpret_cast<const short8*>(
-          p_lds + lc * LST + s * 32 + lg * 8);
-      short8 dsa = *reinterpret_cast<const short8*>(
-          ds_lds + lc * LST + s * 32 + lg * 8);
+      for (int s = 0; s < 2; ++s) {
+        short8 pa = *reinterpret_cast<const short8*>(
+            p_lds + lc * LST + s * 32 + lg * 8);
+        short8 dsa = *reinterpret_cast<const short8*>(
+            ds_lds + lc * LST + s * 32 + lg * 8);
 #pragma unroll
-      for (int t = 0; t < DT; ++t) {
-        short8 dob = *reinterpret_cast<const short8*>(
-            doT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
-        short8 qb = *reinterpret_cast<const short8*>(
-            qT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
-        acc_dv[t] = MFMA(pa, dob, acc_dv[t]);
-        acc_dk[t] = MFMA(dsa, qb, acc_dk[t]);
+        for (int t = 0; t < DT; ++t) {
+          short8 dob = *reinterpret_cast<const short8*>(
+              doT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+          short8 qb = *reinterpret_cast<const short8*>(
+              qT_lds + (t * 16 + lc) * LST + s * 32 + lg * 8);
+          acc_dv[m][t] = MFMA(pa, dob, acc_dv[m][t]);
+          acc_dk[m][t] = MFMA(dsa, qb, acc_dk[m][t]);
+        }
       }
     }
   }
 
-  // store dK/dV rows kv = kv0 + lg*4 + r, col d = t*16+lc.
+  // store dK/dV rows kv = kv0 + m*16 + lg*4 + r, col d = t*16+lc.
   // GQA: emitted per QUERY head; the wrapper group-reduces to kv heads.
   u16* dKp = dk + ((long long)b * S + kv0) * qs + (long long)h * D;
   u16* dVp = dv + ((long long)b * S + kv0) * qs + (long long)h * D;
 #pragma unroll
-  for (int r = 0; r < 4; ++r)
+  for (int m = 0; m < M2; ++m)
 #pragma unroll
-    for (int t = 0; t < DT; ++t) {
-      dKp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
-          f32_to_bf16(acc_dk[t][r]);
-      dVp[(long long)(lg * 4 + r) * qs + t * 16 + lc] =
-          f32_to_bf16(acc_dv[t][r]);
-    }
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int t = 0; t < DT; ++t) {
+        dKp[(long long)(m * 16 + lg * 4 + r) * qs + t * 16 + lc] =
+            f32_to_bf16(acc_dk[m][t][r]);
+        dVp[(long long)(m * 16 + lg * 4 + r) * qs + t * 16 + lc] =
+            f32_to_bf16(acc_dv[m][t][r]);
+      }
 }
 
 }  // namespace
@@ -330,37 +361,27 @@ void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       const void* dO, const float* lse, const float* delta,
                       void* dq, int B, int S, int H, int Hkv, int D,
                       float scale, int window, hipStream_t stream) {
-  dim3 grid(S / TILE, B * H);
+  const bool wide = (S % 128 == 0);
+  dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds = (D * LST + 2 * TILE * (D + 8) + 4 * 16 * LST) * sizeof(u16);
-  if (D == 64)
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<64>, grid, dim3(256), lds, stream,
-                       (const u16*)q, (const u16*)k, (const u16*)v,
-                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
-                       scale, window);
-  else
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<128>, grid, dim3(256), lds, stream,
-                       (const u16*)q, (const u16*)k, (const u16*)v,
-                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
-                       scale, window);
+#define LQ(DD, QQ) hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, QQ>), grid,     dim3(256), lds, stream, (const u16*)q, (const u16*)k, (const u16*)v,     (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv, scale, window)
+  if (D == 64) { if (wide) LQ(64, 32); else LQ(64, 16); }
+  else         { if (wide) LQ(128, 32); else LQ(128, 16); }
+#undef LQ
 }
 
 void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        const void* dO, const float* lse, const float* delta,
                        void* dk, void* dv, int B, int S, int H, int Hkv,
                        int D, float scale, int window, hipStream_t stream) {
-  dim3 grid(S / TILE, B * H);
+  const bool wide = (S % 128 == 0);
+  dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds =
       (2 * D * LST + 2 * TILE * (D + 8) + 8 * 16 * LST) * sizeof(u16);
-  if (D == 64)
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<64>, grid, dim3(256), lds, stream,
-                       (const u16*)q, (const u16*)k, (const u16*)v,
-                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                       Hkv, scale, window);
-  else
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<128>, grid, dim3(256), lds, stream,
-                       (const u16*)q, (const u16*)k, (const u16*)v,
-                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                       Hkv, scale, window);
+#define LK(DD, QQ) hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, QQ>), grid,     dim3(256), lds, stream, (const u16*)q, (const u16*)k, (const u16*)v,     (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H, Hkv, scale, window)
+  if (D == 64) { if (wide) LK(64, 32); else LK(64, 16); }
+  else         { if (wide) LK(128, 32); else LK(128, 16); }
+#undef LK
 }
 
 }  // extern "C"
