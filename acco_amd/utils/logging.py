@@ -37,21 +37,28 @@ class ScalarLogger:
     """JSONL scalar stream, one file per run: the TensorBoard-channel
     equivalent (reference utils/logs_utils.py:187-224)."""
 
-    def __init__(self, log_dir: str, run_name: str, id_run: str, rank: int = 0):
+    def __init__(self, log_dir: str, run_name: str, id_run: str, rank: int = 0,
+                 tensorboard: bool = True):
         self.rank = rank
         self.path: Optional[str] = None
         self._f = None
+        self._tb = None
         if rank == 0:
             d = os.path.join(log_dir, run_name)
             os.makedirs(d, exist_ok=True)
             self.path = os.path.join(d, f"{id_run}.jsonl")
             self._f = open(self.path, "a", buffering=1)
+            if tensorboard:
+                from acco_amd.utils.tb_writer import EventFileWriter
+                self._tb = EventFileWriter(os.path.join(d, "tb", id_run))
 
     def add_scalar(self, tag: str, value: float, step: int) -> None:
         if self._f is None:
             return
         self._f.write(json.dumps({"tag": tag, "value": float(value),
                                   "step": int(step), "ts": time.time()}) + "\n")
+        if self._tb is not None:
+            self._tb.add_scalar(tag, value, step)
 
     def log_training(self, opt_step: int, n_grads: int, rank: int, loss: float,
                      eval_loss: Optional[float], t_beg: float) -> None:
@@ -68,6 +75,9 @@ class ScalarLogger:
         if self._f is not None:
             self._f.close()
             self._f = None
+        if self._tb is not None:
+            self._tb.close()
+            self._tb = None
 
 
 def create_dict_result(dict_args: Dict[str, Any], world_size: int, n_nodes: int,
