@@ -140,6 +140,12 @@ class DecoupledTrainer:
         self.eval_dataset = eval_dataset
         if train_dataset is None:
             return
+        if (isinstance(train_dataset, torch.utils.data.IterableDataset)
+                and getattr(self.args, "group_by_length", False)):
+            # reference prepare_data guard (trainer_base.py:183-191)
+            raise ValueError(
+                "the `--group_by_length` option is only available for "
+                "`Dataset`, not `IterableDataset`")
         is_hf = hasattr(train_dataset, "column_names")
         if is_hf:
             self.train_dataset = train_dataset.shard(
