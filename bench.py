@@ -127,10 +127,12 @@ def main():
     comm = CommBackend(device)
     comm.all_reduce_avg(params)
 
-    from acco_amd.models.fuse import install_fused_projections
-    n_fused = install_fused_projections(model, params, grads)
-    if rank == 0:
-        print(f"# fused projection groups: {n_fused}", file=sys.stderr)
+    if args.method != "ddp":
+        # (DDP keeps per-param GEMMs so its backward-overlap hooks fire)
+        from acco_amd.models.fuse import install_fused_projections
+        n_fused = install_fused_projections(model, params, grads)
+        if rank == 0:
+            print(f"# fused projection groups: {n_fused}", file=sys.stderr)
 
     opt = ShardedAdamW(spec, rank, device, lr=args.lr, betas=(0.9, 0.95),
                        eps=1e-8, weight_decay=0.1)
