@@ -1,7 +1,10 @@
 """Attention kernel micro-benchmark / profiling target (flagship shape)."""
 
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
