@@ -71,6 +71,35 @@ class ShardedAdamW:
         if commit:
             self.step_count += 1
 
+    @torch.no_grad()
+    def consolidate(self, world: int) -> Optional[dict]:
+        """Gather the sharded fp32 state into full flat vectors on rank 0
+        (world-size-independent checkpoints — the capability of torch's
+        ZeroRedundancyOptimizer.consolidate_state_dict which the reference
+        relies on implicitly). Returns None on non-zero ranks."""
+        import torch.distributed as dist
+        spec = self.spec
+        full = {}
+        for name, shard in (("p", self.p), ("m", self.m), ("v", self.v)):
+            if world == 1:
+                gathered = [shard]
+            else:
+                gathered = [torch.empty_like(shard) for _ in range(world)]
+                dist.all_gather(gathered, shard)
+            if self.rank != 0:
+                continue
+            out = torch.empty(spec.total, dtype=torch.float32,
+                              device=shard.device)
+            for r in range(world):
+                for j in range(spec.nb):
+                    spec.seg_view(out, j, r).copy_(
+                        spec.owned_view(gathered[r], j))
+            full[name] = out
+        if self.rank != 0 and world > 1:
+            return None
+        full["step"] = self.step_count
+        return full
+
     def state_dict(self) -> dict:
         return {"step": self.step_count, "p": self.p, "m": self.m, "v": self.v,
                 "lr": self.lr, "beta1": self.beta1, "beta2": self.beta2,

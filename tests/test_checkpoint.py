@@ -43,3 +43,30 @@ def test_scheduler_state_roundtrip():
     t = LRSchedule(1.0, 10, 100)
     t.load_state_dict(s.state_dict())
     assert t.lr() == s.lr()
+
+
+def _worker_consolidate(rank, world, port, tmpdir):
+    import os
+    import torch
+    from tests.dist_utils import init_worker, teardown_worker
+    init_worker(rank, world, port)
+    spec = ShardSpec.build(40, world, buckets=2, align=4)
+    opt = ShardedAdamW(spec, rank, torch.device("cpu"), lr=1e-3)
+    # distinctive per-position values: owned segment holds its global index
+    for j in range(spec.nb):
+        seg0 = j * spec.bucket_elems + rank * spec.seg
+        spec.owned_view(opt.p, j).copy_(
+            torch.arange(seg0, seg0 + spec.seg, dtype=torch.float32))
+    full = opt.consolidate(world)
+    if rank == 0:
+        assert torch.equal(full["p"],
+                           torch.arange(spec.total, dtype=torch.float32))
+        torch.save({"ok": True}, os.path.join(tmpdir, "ok.pt"))
+    teardown_worker()
+
+
+def test_consolidate_ws2():
+    import os
+    from tests.conftest import run_distributed
+    tmpdir = run_distributed(_worker_consolidate, 2, timeout=120)
+    assert os.path.exists(os.path.join(tmpdir, "ok.pt"))
