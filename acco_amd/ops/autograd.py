@@ -141,8 +141,9 @@ class AttentionFn(torch.autograd.Function):
     def backward(ctx, dO):
         q, k, v, o, lse = ctx.saved_tensors
         dO = dO.contiguous()
-        # Delta[b,h,s] = rowsum(dO ∘ O) in fp32, laid out [B,H,S]
-        delta = (dO.float() * o.float()).sum(-1).permute(0, 2, 1).contiguous()
+        # Delta[b,h,s] = rowsum(dO ∘ O) in fp32, laid out [B,H,S] (one
+        # fused kernel; was a 4-kernel ATen cast/mul/reduce chain)
+        delta = ops.hip_ext().attn_delta(dO, o)
         dq, dkq, dvq = ops.hip_ext().attn_bwd(q, k, v, dO, lse, delta,
                                               ctx.scale, ctx.window)
         B, S, H, D = q.shape

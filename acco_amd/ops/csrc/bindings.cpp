@@ -43,6 +43,8 @@ extern "C" void acco_attn_bwd_dq(const void*, const void*, const void*,
                                  const void*, const float*, const float*,
                                  void*, int, int, int, int, int, float, int,
                                  hipStream_t);
+extern "C" void acco_attn_delta(const void*, const void*, float*, long long,
+                                int, int, int, hipStream_t);
 extern "C" void acco_attn_bwd_dkv(const void*, const void*, const void*,
                                   const void*, const float*, const float*,
                                   void*, void*, int, int, int, int, int,
@@ -247,6 +249,17 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {o, lse};
 }
 
+at::Tensor attn_delta(at::Tensor dO, at::Tensor o) {
+  CHECK_BF16_CONTIG(dO); CHECK_BF16_CONTIG(o);
+  const long long B = dO.size(0);
+  const int S = (int)dO.size(1), H = (int)dO.size(2), D = (int)dO.size(3);
+  TORCH_CHECK(D % 8 == 0);
+  auto delta = at::empty({B, H, S}, dO.options().dtype(at::kFloat));
+  acco_attn_delta(dO.data_ptr(), o.data_ptr(), delta.data_ptr<float>(), B, S,
+                  H, D, cur_stream());
+  return delta;
+}
+
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor dO, at::Tensor lse,
                                  at::Tensor delta, double scale,
@@ -290,6 +303,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("attn_fwd", &attn_fwd);
+  m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);
   m.attr("_gfx950") = true;
 }
