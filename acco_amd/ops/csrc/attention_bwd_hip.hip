@@ -138,6 +138,7 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     for (int m = 0; m < M2; ++m) {
       // S^T = K·Q^T and dP^T = V·dO^T (C: col=q=lc, row=kv spread)
       f32x4 st[4], dpt[4];
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int m16 = 0; m16 < 4; ++m16) {
         f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
@@ -153,6 +154,7 @@ void attn_bwd_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
         st[m16] = a1;
         dpt[m16] = a2;
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // dS^T = P^T ∘ (dP^T - Delta) * scale, P = exp(S*scale - lse)
       const int q_g = q0 + m * 16 + lc;
@@ -275,6 +277,7 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     for (int m = 0; m < M2; ++m) {
       // S = Q·K^T, dP = dO·V^T (C: col = kv = lc, row = q spread)
       f32x4 st[4], dpt[4];
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int m16 = 0; m16 < 4; ++m16) {
         f32x4 a1 = {0.f, 0.f, 0.f, 0.f}, a2 = {0.f, 0.f, 0.f, 0.f};
@@ -290,6 +293,7 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
         st[m16] = a1;
         dpt[m16] = a2;
       }
+      __builtin_amdgcn_s_setprio(0);
 
       const int kv_g = kv0 + m * 16 + lc;
 #pragma unroll
@@ -361,12 +365,12 @@ void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       const void* dO, const float* lse, const float* delta,
                       void* dq, int B, int S, int H, int Hkv, int D,
                       float scale, int window, hipStream_t stream) {
-  const bool wide = (S % 128 == 0);
+  const bool wide = (S % 128 == 0) && (D == 64);
   dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds = (D * LST + 2 * TILE * (D + 8) + 4 * 16 * LST) * sizeof(u16);
 #define LQ(DD, QQ) hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, QQ>), grid,     dim3(256), lds, stream, (const u16*)q, (const u16*)k, (const u16*)v,     (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv, scale, window)
   if (D == 64) { if (wide) LQ(64, 32); else LQ(64, 16); }
-  else         { if (wide) LQ(128, 32); else LQ(128, 16); }
+  else         LQ(128, 16);
 #undef LQ
 }
 
@@ -374,13 +378,13 @@ void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        const void* dO, const float* lse, const float* delta,
                        void* dk, void* dv, int B, int S, int H, int Hkv,
                        int D, float scale, int window, hipStream_t stream) {
-  const bool wide = (S % 128 == 0);
+  const bool wide = (S % 128 == 0) && (D == 64);
   dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds =
       (2 * D * LST + 2 * TILE * (D + 8) + 8 * 16 * LST) * sizeof(u16);
 #define LK(DD, QQ) hipLaunchKernelGGL((attn_bwd_dkv_kernel<DD, QQ>), grid,     dim3(256), lds, stream, (const u16*)q, (const u16*)k, (const u16*)v,     (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H, Hkv, scale, window)
   if (D == 64) { if (wide) LK(64, 32); else LK(64, 16); }
-  else         { if (wide) LK(128, 32); else LK(128, 16); }
+  else         LK(128, 16);
 #undef LK
 }
 

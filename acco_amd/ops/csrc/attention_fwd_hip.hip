@@ -128,6 +128,7 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     for (int m = 0; m < M2; ++m) {
       // ---- S^T: st[m16] = K_sub · Q^T (C: col=q=lc, row=kv spread)
       f32x4 st[4];
+      __builtin_amdgcn_s_setprio(1);   // guide T5: favor the MFMA cluster
 #pragma unroll
       for (int m16 = 0; m16 < 4; ++m16) {
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -139,6 +140,7 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
         }
         st[m16] = acc;
       }
+      __builtin_amdgcn_s_setprio(0);
 
       // ---- mask + online softmax (per col q=lc)
       const int q_g = q0 + m * 16 + lc;
@@ -194,9 +196,10 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 #pragma unroll
         for (int r = 0; r < 4; ++r) acc_o[m][t][r] *= alpha_row[r];
 
-      __builtin_amdgcn_s_waitcnt(0);   // lgkm: own-wave P writes
-
+      // (plain LDS accesses: hipcc's counted lgkmcnt orders the P
+      // write->read pair; no full-counter drain needed)
       // ---- PV: A = P[q=lc][kv], B = V^T from LDS
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 2; ++s) {
         short8 pa = *reinterpret_cast<const short8*>(
@@ -208,6 +211,7 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
           acc_o[m][t] = MFMA(pa, vb, acc_o[m][t]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
   }
 
@@ -241,13 +245,14 @@ extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               hipStream_t stream) {
   const int lds_bytes =
       (D * LSTRIDE + KT * (D + 8) + 4 * 16 * LSTRIDE) * sizeof(u16);
-  const bool wide = (S % 128 == 0);
+  // D=128 at QW=32 needs ~197 VGPR -> 1 wave/SIMD; keep QW=16 there
+  const bool wide = (S % 128 == 0) && (D == 64);
   const int qt_rows = wide ? 128 : 64;
   dim3 grid(S / qt_rows, B * H);
 #define LA(DD, QQ) hipLaunchKernelGGL((attn_fwd_kernel<DD, QQ>), grid, \
     dim3(256), lds_bytes, stream, (const u16*)q, (const u16*)k, \
     (const u16*)v, (u16*)o, lse, S, H, Hkv, scale, window)
   if (D == 64) { if (wide) LA(64, 32); else LA(64, 16); }
-  else         { if (wide) LA(128, 32); else LA(128, 16); }
+  else         LA(128, 16);
 #undef LA
 }
