@@ -11,7 +11,7 @@ target the arena directly with zero staging copies.
 
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.nn as nn

@@ -13,7 +13,7 @@ from __future__ import annotations
 import contextlib
 import os
 import time
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 from torch.utils.data import DataLoader, Dataset, RandomSampler

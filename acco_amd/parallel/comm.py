@@ -24,7 +24,7 @@ to the algorithm: the fp32 optimizer shard is bucket-major contiguous).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional
+
 
 import torch
 import torch.distributed as dist

@@ -2,8 +2,6 @@
 causal, GQA, local window. Asymmetric random data (guide: symmetric inputs
 mask transposed-layout bugs)."""
 
-import math
-
 import pytest
 import torch
 
