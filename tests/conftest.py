@@ -40,16 +40,22 @@ def run_distributed(worker, world_size: int, args=(), timeout: float = 180.0):
 
     import time
 
-    port = free_port()
-    tmpdir = tempfile.mkdtemp(prefix="acco_test_")
-    ctx = mp.spawn(worker, args=(world_size, port, tmpdir) + tuple(args),
-                   nprocs=world_size, join=False)
-    deadline = time.time() + timeout
-    # ctx.join(t) returns False whenever *some* process is still alive after
-    # one wait round — poll until the deadline.
-    while not ctx.join(timeout=5):
-        if time.time() > deadline:
-            for p in ctx.processes:
-                p.terminate()
-            raise RuntimeError("distributed test timed out")
-    return tmpdir
+    last_exc = None
+    for attempt in range(2):      # one retry absorbs port/rendezvous flakes
+        port = free_port()
+        tmpdir = tempfile.mkdtemp(prefix="acco_test_")
+        try:
+            ctx = mp.spawn(worker, args=(world_size, port, tmpdir) + tuple(args),
+                           nprocs=world_size, join=False)
+            deadline = time.time() + timeout
+            # ctx.join(t) returns False whenever *some* process is still
+            # alive after one wait round — poll until the deadline.
+            while not ctx.join(timeout=5):
+                if time.time() > deadline:
+                    for p in ctx.processes:
+                        p.terminate()
+                    raise RuntimeError("distributed test timed out")
+            return tmpdir
+        except RuntimeError as e:
+            last_exc = e
+    raise last_exc
