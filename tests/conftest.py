@@ -56,6 +56,6 @@ def run_distributed(worker, world_size: int, args=(), timeout: float = 180.0):
                         p.terminate()
                     raise RuntimeError("distributed test timed out")
             return tmpdir
-        except RuntimeError as e:
+        except Exception as e:   # ProcessRaisedException is not a RuntimeError
             last_exc = e
     raise last_exc
