@@ -82,3 +82,49 @@ def test_trainer_acco_ws2():
 def test_trainer_dpu_ws2():
     tmpdir = run_distributed(_worker_dpu, 2, timeout=300)
     _check(tmpdir, "dpu")
+
+
+def _worker_acco_bf16(rank, world, port, tmpdir):
+    """bf16 arenas + fp32 master shard on CPU (the mixed-precision data
+    path of the GPU configuration, minus autocast)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.chdir(tmpdir)
+
+    from acco_amd.config import load_config
+    from acco_amd.data.synthetic import SyntheticCausalLMDataset
+    from acco_amd.engine.trainer import DecoupledTrainer
+    from acco_amd.models import GPTNeoConfig, GPTNeoForCausalLM
+
+    cfg = load_config([
+        "train=acco", "train.nb_steps_tot=6", "train.batch_size=2",
+        "train.max_length=16", "train.use_mixed_precision=true",
+        "train.save=false", "train.dataloader_num_workers=0",
+        "train.comm_buckets=2", "train.dataloader_persistent_workers=false",
+    ])
+    torch.manual_seed(42)
+    mcfg = GPTNeoConfig(hidden_size=32, num_layers=1, num_heads=2,
+                        vocab_size=64, max_position_embeddings=32,
+                        window_size=8)
+    model = GPTNeoForCausalLM(mcfg)
+    ds = SyntheticCausalLMDataset(16, 16, 64, seed=5 + rank)
+    trainer = DecoupledTrainer(model=model, train_dataset=ds,
+                               eval_dataset=None, args=cfg.train,
+                               run_name="bf16")
+    assert trainer.params.dtype == torch.bfloat16
+    assert trainer.opt.p.dtype == torch.float32
+    trainer.train()
+    flat = trainer.params[:trainer.n_live].float().clone()
+    torch.save({"params": flat}, os.path.join(tmpdir, f"p_bf16_{rank}.pt"))
+    teardown_worker()
+
+
+def test_trainer_acco_bf16_ws2():
+    tmpdir = run_distributed(_worker_acco_bf16, 2, timeout=300)
+    res = [torch.load(os.path.join(tmpdir, f"p_bf16_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert torch.isfinite(res[0]["params"]).all()
