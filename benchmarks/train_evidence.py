@@ -59,7 +59,7 @@ def main():
     first = sum(l for _, l in losses[:5]) / 5
     last = sum(l for _, l in losses[-5:]) / 5
     print(f"first5={first:.3f} last5={last:.3f}")
-    assert last < 0.6 * first, "training did not learn"
+    assert last < first - 1.0, "training did not learn (expect >1 nat drop)"
     print("TRAIN_EVIDENCE_OK")
 
 
