@@ -238,10 +238,20 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 
 }  // namespace
 
+extern "C" void acco_attn_fwd32(const void* q, const void* k, const void* v,
+                                void* o, float* lse, int B, int S, int H,
+                                int Hkv, int D, float scale, int window,
+                                hipStream_t stream);
+
 extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               void* o, float* lse, int B, int S, int H,
                               int Hkv, int D, float scale, int window,
                               hipStream_t stream) {
+  if (D == 64 && S % 256 == 0) {
+    // v4: 32x32 MFMA + in-register softmax (attention_fwd32.hip)
+    acco_attn_fwd32(q, k, v, o, lse, B, S, H, Hkv, D, scale, window, stream);
+    return;
+  }
   const int lds_bytes =
       (D * LSTRIDE + KT * (D + 8) + 4 * 16 * LSTRIDE) * sizeof(u16);
   // D=128 at QW=32 needs ~197 VGPR -> 1 wave/SIMD; keep QW=16 there
