@@ -359,12 +359,26 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 
 }  // namespace
 
+extern "C" void acco_attn_bwd32_dq(const void*, const void*, const void*,
+                                   const void*, const float*, const float*,
+                                   void*, int, int, int, int, int, float,
+                                   int, hipStream_t);
+extern "C" void acco_attn_bwd32_dkv(const void*, const void*, const void*,
+                                    const void*, const float*, const float*,
+                                    void*, void*, int, int, int, int, int,
+                                    float, int, hipStream_t);
+
 extern "C" {
 
 void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       const void* dO, const float* lse, const float* delta,
                       void* dq, int B, int S, int H, int Hkv, int D,
                       float scale, int window, hipStream_t stream) {
+  if (D == 64 && S % 256 == 0) {
+    acco_attn_bwd32_dq(q, k, v, dO, lse, delta, dq, B, S, H, Hkv, D, scale,
+                       window, stream);
+    return;
+  }
   const bool wide = (S % 128 == 0) && (D == 64);
   dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds = (D * LST + 2 * TILE * (D + 8) + 4 * 16 * LST) * sizeof(u16);
@@ -378,6 +392,11 @@ void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        const void* dO, const float* lse, const float* delta,
                        void* dk, void* dv, int B, int S, int H, int Hkv,
                        int D, float scale, int window, hipStream_t stream) {
+  if (D == 64 && S % 256 == 0) {
+    acco_attn_bwd32_dkv(q, k, v, dO, lse, delta, dk, dv, B, S, H, Hkv, D,
+                        scale, window, stream);
+    return;
+  }
   const bool wide = (S % 128 == 0) && (D == 64);
   dim3 grid(S / (wide ? 128 : 64), B * H);
   const int lds =
