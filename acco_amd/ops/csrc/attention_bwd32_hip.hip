@@ -345,10 +345,16 @@ void acco_attn_bwd32_dq(const void* q, const void* k, const void* v,
                         float scale, int window, hipStream_t stream) {
   dim3 grid(S / BT, B * H);
   const int lds = (2 * KT * (D + 8) + D * LST) * sizeof(u16);
-  hipLaunchKernelGGL(attn_bwd32_dq_kernel<64>, grid, dim3(512), lds, stream,
-                     (const u16*)q, (const u16*)k, (const u16*)v,
-                     (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv, scale,
-                     window);
+  if (D == 64)
+    hipLaunchKernelGGL(attn_bwd32_dq_kernel<64>, grid, dim3(512), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
+                       scale, window);
+  else
+    hipLaunchKernelGGL(attn_bwd32_dq_kernel<128>, grid, dim3(512), lds,
+                       stream, (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
+                       scale, window);
 }
 
 void acco_attn_bwd32_dkv(const void* q, const void* k, const void* v,

@@ -374,7 +374,7 @@ void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       const void* dO, const float* lse, const float* delta,
                       void* dq, int B, int S, int H, int Hkv, int D,
                       float scale, int window, hipStream_t stream) {
-  if (D == 64 && S % 256 == 0) {
+  if ((D == 64 || D == 128) && S % 256 == 0) {
     acco_attn_bwd32_dq(q, k, v, dO, lse, delta, dq, B, S, H, Hkv, D, scale,
                        window, stream);
     return;

@@ -248,7 +248,12 @@ extern "C" void acco_attn_fwd32(const void* q, const void* k, const void* v,
                                 hipStream_t stream) {
   dim3 grid(S / QT, B * H);
   const int lds = (KT * (D + 8) + D * LST) * sizeof(u16);
-  hipLaunchKernelGGL(attn_fwd32_kernel<64>, grid, dim3(512), lds, stream,
-                     (const u16*)q, (const u16*)k, (const u16*)v, (u16*)o,
-                     lse, S, H, Hkv, scale, window);
+  if (D == 64)
+    hipLaunchKernelGGL(attn_fwd32_kernel<64>, grid, dim3(512), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v, (u16*)o,
+                       lse, S, H, Hkv, scale, window);
+  else
+    hipLaunchKernelGGL(attn_fwd32_kernel<128>, grid, dim3(512), lds, stream,
+                       (const u16*)q, (const u16*)k, (const u16*)v, (u16*)o,
+                       lse, S, H, Hkv, scale, window);
 }

@@ -248,7 +248,7 @@ extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               void* o, float* lse, int B, int S, int H,
                               int Hkv, int D, float scale, int window,
                               hipStream_t stream) {
-  if (D == 64 && S % 256 == 0) {
+  if ((D == 64 || D == 128) && S % 256 == 0) {
     // v4: 32x32 MFMA + in-register softmax (attention_fwd32.hip)
     acco_attn_fwd32(q, k, v, o, lse, B, S, H, Hkv, D, scale, window, stream);
     return;

@@ -54,6 +54,15 @@ def test_attn_d128():
     run_case(B=1, S=128, H=4, Hkv=4, D=128, seed=2)
 
 
+def test_attn_d128_v4_path():
+    # S % 256 == 0 → the 32x32 v4 fwd/dq kernels with the v3 dkv
+    run_case(B=1, S=256, H=4, Hkv=2, D=128, seed=7)
+
+
+def test_attn_d64_v4_path_gqa_window():
+    run_case(B=1, S=512, H=8, Hkv=2, D=64, window=96, seed=8)
+
+
 def test_attn_local_window():
     # GPT-Neo-style: no scaling (1.0) + banded window; small magnitudes so
     # unscaled scores stay sane
