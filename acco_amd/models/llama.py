@@ -78,6 +78,9 @@ class LlamaMLP(nn.Module):
             from acco_amd.models.fuse import FusedArenaLinearFn
             w, g, splits = self._fused_gate_up
             gu = FusedArenaLinearFn.apply(x, w, g)
+            if gu.is_cuda and ops.have_kernel("swiglu_packed_fwd"):
+                from acco_amd.ops.autograd import SwiGLUPackedFn
+                return self.down_proj(SwiGLUPackedFn.apply(gu))
             gate, up = torch.split(gu, splits, dim=-1)
             return self.down_proj(ops.swiglu(gate.contiguous(),
                                              up.contiguous()))

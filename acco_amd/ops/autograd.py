@@ -28,6 +28,22 @@ class SwiGLUFn(torch.autograd.Function):
         return dg, du
 
 
+class SwiGLUPackedFn(torch.autograd.Function):
+    """SwiGLU over the packed [.., 2I] fused gate_up output: no
+    split/contiguous forward, single dgu write backward (no cat)."""
+
+    @staticmethod
+    def forward(ctx, gu):
+        gu = gu.contiguous()
+        ctx.save_for_backward(gu)
+        return ops.hip_ext().swiglu_packed_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dout):
+        (gu,) = ctx.saved_tensors
+        return ops.hip_ext().swiglu_packed_bwd(dout.contiguous(), gu)
+
+
 class GeluNewFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):

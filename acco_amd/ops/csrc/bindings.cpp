@@ -11,9 +11,10 @@ extern "C" void acco_fused_adamw_launch(
     float scale, float lr, float beta1, float beta2, float eps,
     float weight_decay, long long step_plus_1, hipStream_t stream);
 extern "C" void acco_swiglu_fwd(const void*, const void*, void*, long long,
-                                hipStream_t);
+                                int, long long, hipStream_t);
 extern "C" void acco_swiglu_bwd(const void*, const void*, const void*, void*,
-                                void*, long long, hipStream_t);
+                                void*, long long, int, long long,
+                                hipStream_t);
 extern "C" void acco_gelu_fwd(const void*, void*, long long, hipStream_t);
 extern "C" void acco_gelu_bwd(const void*, const void*, void*, long long,
                               hipStream_t);
@@ -51,6 +52,8 @@ extern "C" void acco_attn_bwd_dkv(const void*, const void*, const void*,
                                   float, int, hipStream_t);
 
 namespace {
+
+using u16 = unsigned short;
 
 void fused_adamw(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                  int64_t step, double lr, double beta1, double beta2,
@@ -94,19 +97,51 @@ static hipStream_t cur_stream() {
 at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u) {
   CHECK_BF16_CONTIG(g); CHECK_BF16_CONTIG(u);
   TORCH_CHECK(g.numel() == u.numel() && g.numel() % 8 == 0);
+  const int I = (int)g.size(-1);
+  TORCH_CHECK(I % 8 == 0);
   auto out = at::empty_like(g);
   acco_swiglu_fwd(g.data_ptr(), u.data_ptr(), out.data_ptr(), g.numel(),
-                  cur_stream());
+                  I, I / 8, cur_stream());
   return out;
 }
 
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor g, at::Tensor u) {
   CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(g); CHECK_BF16_CONTIG(u);
+  const int I = (int)g.size(-1);
   auto dg = at::empty_like(g);
   auto du = at::empty_like(u);
   acco_swiglu_bwd(dout.data_ptr(), g.data_ptr(), u.data_ptr(), dg.data_ptr(),
-                  du.data_ptr(), g.numel(), cur_stream());
+                  du.data_ptr(), g.numel(), I, I / 8, cur_stream());
   return {dg, du};
+}
+
+// SwiGLU over the packed [rows, 2I] fused gate_up projection output:
+// out = silu(gu[:, :I]) * gu[:, I:]; backward emits dgu in one pass
+// (no torch.split → no contiguous copies forward, no cat backward).
+at::Tensor swiglu_packed_fwd(at::Tensor gu) {
+  CHECK_BF16_CONTIG(gu);
+  const int twoI = (int)gu.size(-1);
+  TORCH_CHECK(twoI % 16 == 0);
+  const int I = twoI / 2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = at::empty(sizes, gu.options());
+  const u16* base = (const u16*)gu.data_ptr();
+  acco_swiglu_fwd(base, base + I, out.data_ptr(), gu.numel() / 2, I,
+                  twoI / 8, cur_stream());
+  return out;
+}
+
+at::Tensor swiglu_packed_bwd(at::Tensor dout, at::Tensor gu) {
+  CHECK_BF16_CONTIG(dout); CHECK_BF16_CONTIG(gu);
+  const int twoI = (int)gu.size(-1);
+  const int I = twoI / 2;
+  auto dgu = at::empty_like(gu);
+  const u16* base = (const u16*)gu.data_ptr();
+  u16* dbase = (u16*)dgu.data_ptr();
+  acco_swiglu_bwd(dout.data_ptr(), base, base + I, dbase, dbase + I,
+                  gu.numel() / 2, I, twoI / 8, cur_stream());
+  return dgu;
 }
 
 at::Tensor gelu_fwd(at::Tensor x) {
@@ -293,6 +328,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "commit=false = ACCO tentative step");
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("swiglu_packed_fwd", &swiglu_packed_fwd);
+  m.def("swiglu_packed_bwd", &swiglu_packed_bwd);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);

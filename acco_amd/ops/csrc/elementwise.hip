@@ -29,16 +29,21 @@ ACCO_DEV float gelu_new_grad_f(float x) {
 using u16 = unsigned short;
 
 // ---- SwiGLU: out = silu(g) * u
+// g/u may be row-strided sections of one packed [rows][2I] tensor
+// (the fused gate_up projection output): row = i8 / (I/8) vec8 groups.
 __global__ void swiglu_fwd_kernel(const u16* __restrict__ g,
                                   const u16* __restrict__ u,
-                                  u16* __restrict__ out, long long n8) {
+                                  u16* __restrict__ out, long long n8,
+                                  int i8_per_row, long long row_stride) {
   const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = i0; i < n8; i += stride) {
-    ushort4 gv0 = reinterpret_cast<const ushort4*>(g)[2 * i];
-    ushort4 gv1 = reinterpret_cast<const ushort4*>(g)[2 * i + 1];
-    ushort4 uv0 = reinterpret_cast<const ushort4*>(u)[2 * i];
-    ushort4 uv1 = reinterpret_cast<const ushort4*>(u)[2 * i + 1];
+    const long long row = i / i8_per_row;
+    const long long co = (i % i8_per_row) + row * row_stride;
+    ushort4 gv0 = reinterpret_cast<const ushort4*>(g)[2 * co];
+    ushort4 gv1 = reinterpret_cast<const ushort4*>(g)[2 * co + 1];
+    ushort4 uv0 = reinterpret_cast<const ushort4*>(u)[2 * co];
+    ushort4 uv1 = reinterpret_cast<const ushort4*>(u)[2 * co + 1];
     u16 gs[8] = {gv0.x, gv0.y, gv0.z, gv0.w, gv1.x, gv1.y, gv1.z, gv1.w};
     u16 us[8] = {uv0.x, uv0.y, uv0.z, uv0.w, uv1.x, uv1.y, uv1.z, uv1.w};
     u16 os[8];
@@ -54,19 +59,22 @@ __global__ void swiglu_fwd_kernel(const u16* __restrict__ g,
 }
 
 // dgate = dout * u * d/dg[g*sig(g)]; dup = dout * silu(g)
+// (dout contiguous [rows][I]; g/u/dg/du row-strided packed sections)
 __global__ void swiglu_bwd_kernel(const u16* __restrict__ dout,
                                   const u16* __restrict__ g,
                                   const u16* __restrict__ u,
                                   u16* __restrict__ dg,
-                                  u16* __restrict__ du, long long n8) {
+                                  u16* __restrict__ du, long long n8,
+                                  int i8_per_row, long long row_stride) {
   const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = i0; i < n8; i += stride) {
+    const long long co = (i % i8_per_row) + (i / i8_per_row) * row_stride;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       ushort4 dv = reinterpret_cast<const ushort4*>(dout)[2 * i + h];
-      ushort4 gv = reinterpret_cast<const ushort4*>(g)[2 * i + h];
-      ushort4 uv = reinterpret_cast<const ushort4*>(u)[2 * i + h];
+      ushort4 gv = reinterpret_cast<const ushort4*>(g)[2 * co + h];
+      ushort4 uv = reinterpret_cast<const ushort4*>(u)[2 * co + h];
       u16 ds[4] = {dv.x, dv.y, dv.z, dv.w};
       u16 gs[4] = {gv.x, gv.y, gv.z, gv.w};
       u16 us[4] = {uv.x, uv.y, uv.z, uv.w};
@@ -82,8 +90,8 @@ __global__ void swiglu_bwd_kernel(const u16* __restrict__ dout,
         dgo[k] = f32_to_bf16(df * uf * dsilu);
         duo[k] = f32_to_bf16(df * silu);
       }
-      reinterpret_cast<ushort4*>(dg)[2 * i + h] = make_ushort4(dgo[0], dgo[1], dgo[2], dgo[3]);
-      reinterpret_cast<ushort4*>(du)[2 * i + h] = make_ushort4(duo[0], duo[1], duo[2], duo[3]);
+      reinterpret_cast<ushort4*>(dg)[2 * co + h] = make_ushort4(dgo[0], dgo[1], dgo[2], dgo[3]);
+      reinterpret_cast<ushort4*>(du)[2 * co + h] = make_ushort4(duo[0], duo[1], duo[2], duo[3]);
     }
   }
 }
@@ -134,20 +142,22 @@ __global__ void gelu_bwd_kernel(const u16* __restrict__ dout,
 extern "C" {
 
 void acco_swiglu_fwd(const void* g, const void* u, void* out, long long n,
-                     hipStream_t s) {
+                     int cols, long long row_stride8, hipStream_t s) {
   long long n8 = n / 8;
   int grid = elementwise_grid(n8, 256);
   hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid), dim3(256), 0, s,
-                     (const u16*)g, (const u16*)u, (u16*)out, n8);
+                     (const u16*)g, (const u16*)u, (u16*)out, n8, cols / 8,
+                     row_stride8);
 }
 
 void acco_swiglu_bwd(const void* dout, const void* g, const void* u, void* dg,
-                     void* du, long long n, hipStream_t s) {
+                     void* du, long long n, int cols, long long row_stride8,
+                     hipStream_t s) {
   long long n8 = n / 8;
   int grid = elementwise_grid(n8, 256);
   hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid), dim3(256), 0, s,
                      (const u16*)dout, (const u16*)g, (const u16*)u, (u16*)dg,
-                     (u16*)du, n8);
+                     (u16*)du, n8, cols / 8, row_stride8);
 }
 
 void acco_gelu_fwd(const void* x, void* out, long long n, hipStream_t s) {

@@ -133,3 +133,21 @@ def test_model_ops_route_to_hip():
     for k in ["swiglu_fwd", "gelu_fwd", "rmsnorm_fwd", "layernorm_fwd",
               "rope_fwd", "ce_fwd", "fused_adamw"]:
         assert ops.have_kernel(k), k
+
+
+def test_swiglu_packed_fwd_bwd():
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    from acco_amd.ops.autograd import SwiGLUPackedFn
+    torch.manual_seed(4)
+    gu = torch.randn(3, 64, 512, device="cuda").bfloat16().requires_grad_(True)
+    out = SwiGLUPackedFn.apply(gu)
+    I = 256
+    ref = torch_ref.swiglu(gu.detach().float()[..., :I],
+                           gu.detach().float()[..., I:])
+    assert _close_bf16(out, ref)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    gu32 = gu.detach().float().requires_grad_(True)
+    torch_ref.swiglu(gu32[..., :I], gu32[..., I:]).backward(dout.float())
+    assert _close_bf16(gu.grad, gu32.grad)
