@@ -52,6 +52,15 @@ class LlamaAttention(nn.Module):
             from acco_amd.models.fuse import FusedArenaLinearFn
             w, g, splits = self._fused_qkv
             qkv = FusedArenaLinearFn.apply(x, w, g)
+            if (qkv.is_cuda and cfg.head_dim == 64 and S % 256 == 0
+                    and ops.have_kernel("attn_fwd_packed")):
+                # packed core: RoPE + flash attention directly on the fused
+                # projection output, packed grad back — zero split/cat
+                from acco_amd.ops.autograd import AttnQKVPackedFn
+                o = AttnQKVPackedFn.apply(qkv, cos, sin, cfg.num_heads,
+                                          cfg.num_kv_heads, cfg.head_dim,
+                                          cfg.head_dim ** -0.5)
+                return self.o_proj(o)
             q, k, v = torch.split(qkv, splits, dim=-1)
             q = q.contiguous().view(B, S, cfg.num_heads, cfg.head_dim)
             k = k.contiguous().view(B, S, cfg.num_kv_heads, cfg.head_dim)

@@ -132,6 +132,55 @@ class CausalLMLossFn(torch.autograd.Function):
         return dlogits, None
 
 
+class AttnQKVPackedFn(torch.autograd.Function):
+    """The whole Llama attention core over the PACKED fused-QKV projection
+    output [B, S, (H+2Hkv)·D]: RoPE (strided kernel, q/k sections) +
+    flash attention v4 forward; backward emits one packed dqkv — no
+    torch.split forward copies, no cat backward. D=64, S%256==0."""
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, H, Hkv, D, scale):
+        ext = ops.hip_ext()
+        qkv = qkv.contiguous()
+        cos = cos.float().contiguous()
+        sin = sin.float().contiguous()
+        roped = torch.empty_like(qkv)
+        ext.rope_packed(qkv, roped, cos, sin, 0, H, D, False)
+        ext.rope_packed(qkv, roped, cos, sin, H * D, Hkv, D, False)
+        v_off = (H + Hkv) * D
+        roped[..., v_off:].copy_(qkv[..., v_off:])
+        o, lse = ext.attn_fwd_packed(roped, H, Hkv, D, float(scale), 0)
+        ctx.save_for_backward(roped, o, lse, cos, sin)
+        ctx.meta = (H, Hkv, D, float(scale))
+        return o                     # [B, S, H*D]
+
+    @staticmethod
+    def backward(ctx, dO):
+        ext = ops.hip_ext()
+        roped, o, lse, cos, sin = ctx.saved_tensors
+        H, Hkv, D, scale = ctx.meta
+        B, S, W = roped.shape
+        dO = dO.contiguous()
+        delta = ext.attn_delta(dO.view(B, S, H, D), o.view(B, S, H, D))
+        dqkv = torch.empty_like(roped)
+        dkq, dvq = ext.attn_bwd_packed(roped, dO, lse, delta, dqkv, H, Hkv,
+                                       D, scale, 0)
+        rep = H // Hkv
+        k_off, v_off = H * D, (H + Hkv) * D
+        if rep > 1:
+            dk = dkq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
+            dv = dvq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
+            dqkv[..., k_off:v_off] = dk.reshape(B, S, Hkv * D).bfloat16()
+            dqkv[..., v_off:] = dv.reshape(B, S, Hkv * D).bfloat16()
+        else:
+            dqkv[..., k_off:v_off] = dkq.reshape(B, S, Hkv * D)
+            dqkv[..., v_off:] = dvq.reshape(B, S, Hkv * D)
+        # inverse rotation in place on the q/k grad sections
+        ext.rope_packed(dqkv, dqkv, cos, sin, 0, H, D, True)
+        ext.rope_packed(dqkv, dqkv, cos, sin, k_off, Hkv, D, True)
+        return dqkv, None, None, None, None, None, None
+
+
 class AttentionFn(torch.autograd.Function):
     """Flash-style causal attention, gfx950 MFMA (fwd: online softmax;
     bwd: FA2-style recompute, dq + dkv kernels). [B,S,H,D] layout,

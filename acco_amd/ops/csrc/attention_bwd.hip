@@ -361,11 +361,13 @@ void attn_bwd_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 extern "C" void acco_attn_bwd32_dq(const void*, const void*, const void*,
                                    const void*, const float*, const float*,
                                    void*, int, int, int, int, int, float,
-                                   int, hipStream_t);
+                                   int, long long, long long, long long,
+                                   long long, hipStream_t);
 extern "C" void acco_attn_bwd32_dkv(const void*, const void*, const void*,
                                     const void*, const float*, const float*,
                                     void*, void*, int, int, int, int, int,
-                                    float, int, hipStream_t);
+                                    float, int, long long, long long,
+                                    long long, hipStream_t);
 
 extern "C" {
 
@@ -375,7 +377,8 @@ void acco_attn_bwd_dq(const void* q, const void* k, const void* v,
                       float scale, int window, hipStream_t stream) {
   if ((D == 64 || D == 128) && S % 256 == 0) {
     acco_attn_bwd32_dq(q, k, v, dO, lse, delta, dq, B, S, H, Hkv, D, scale,
-                       window, stream);
+                       window, (long long)H * D, (long long)Hkv * D,
+                       (long long)H * D, (long long)H * D, stream);
     return;
   }
   const bool wide = (S % 128 == 0) && (D == 64);
@@ -393,7 +396,8 @@ void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        int D, float scale, int window, hipStream_t stream) {
   if (D == 64 && S % 256 == 0) {
     acco_attn_bwd32_dkv(q, k, v, dO, lse, delta, dk, dv, B, S, H, Hkv, D,
-                        scale, window, stream);
+                        scale, window, (long long)H * D, (long long)Hkv * D,
+                        (long long)H * D, stream);
     return;
   }
   const bool wide = (S % 128 == 0) && (D == 64);

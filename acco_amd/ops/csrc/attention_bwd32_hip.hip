@@ -91,7 +91,9 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                           const float* __restrict__ lse,
                           const float* __restrict__ delta,
                           u16* __restrict__ dq,
-                          int S, int H, int Hkv, float scale, int window) {
+                          int S, int H, int Hkv, float scale, int window,
+                          long long q_rs, long long kv_rs, long long do_rs,
+                          long long dq_rs) {
   constexpr int KS = D / 16;
   constexpr int DT = D / 32;
   constexpr int KROW = D + 8;
@@ -105,10 +107,10 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   u16* v_row = k_row + KT * KROW;          // [KT][KROW]
   u16* kT_lds = v_row + KT * KROW;         // [D][LST]
 
-  const long long qs = (long long)H * D, ks = (long long)Hkv * D;
+  const long long qs = q_rs, ks = kv_rs;
   const int q0 = qt * BT + wave * QW;
   const u16* Qp = q + ((long long)b * S + q0) * qs + (long long)h * D;
-  const u16* dOp = dO + ((long long)b * S + q0) * qs + (long long)h * D;
+  const u16* dOp = dO + ((long long)b * S + q0) * do_rs + (long long)h * D;
   const u16* Kb = k + (long long)b * S * ks + (long long)hkv * D;
   const u16* Vb = v + (long long)b * S * ks + (long long)hkv * D;
 
@@ -118,7 +120,7 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     qf[s] = *reinterpret_cast<const short8*>(
         Qp + (long long)lq * qs + s * 16 + hi * 8);
     dof[s] = *reinterpret_cast<const short8*>(
-        dOp + (long long)lq * qs + s * 16 + hi * 8);
+        dOp + (long long)lq * do_rs + s * 16 + hi * 8);
   }
   const float lse_c = lse[(long long)bh * S + q0 + lq];
   const float delta_c = delta[(long long)bh * S + q0 + lq];
@@ -190,13 +192,13 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
   }
 
-  u16* dQp = dq + ((long long)b * S + q0) * qs + (long long)h * D;
+  u16* dQp = dq + ((long long)b * S + q0) * dq_rs + (long long)h * D;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
 #pragma unroll
     for (int t = 0; t < DT; ++t)
-      dQp[(long long)qrow * qs + t * 32 + lq] = f32_to_bf16(acc_dq[t][r]);
+      dQp[(long long)qrow * dq_rs + t * 32 + lq] = f32_to_bf16(acc_dq[t][r]);
   }
 }
 
@@ -209,7 +211,8 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                            const float* __restrict__ lse,
                            const float* __restrict__ delta,
                            u16* __restrict__ dk, u16* __restrict__ dv,
-                           int S, int H, int Hkv, float scale, int window) {
+                           int S, int H, int Hkv, float scale, int window,
+                           long long q_rs, long long kv_rs, long long do_rs) {
   constexpr int KS = D / 16;
   constexpr int DT = D / 32;
   constexpr int KROW = D + 8;
@@ -224,12 +227,13 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   u16* qT_lds = do_row + KT * KROW;         // [D][LST]
   u16* doT_lds = qT_lds + D * LST;          // [D][LST]
 
-  const long long qs = (long long)H * D, ks = (long long)Hkv * D;
+  const long long qs = q_rs, ks = kv_rs;
+  const long long od = (long long)H * D;   // dk/dv temps: contiguous [B,S,H,D]
   const int kv0 = jb * BT + wave * QW;
   const u16* Kp = k + ((long long)b * S + kv0) * ks + (long long)hkv * D;
   const u16* Vp = v + ((long long)b * S + kv0) * ks + (long long)hkv * D;
   const u16* Qb = q + (long long)b * S * qs + (long long)h * D;
-  const u16* dOb = dO + (long long)b * S * qs + (long long)h * D;
+  const u16* dOb = dO + (long long)b * S * do_rs + (long long)h * D;
 
   // K^T / V^T as B operands: lane = K[kv=lq][d=hi*8+i+16s]
   short8 kTf[KS], vTf[KS];
@@ -258,9 +262,9 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   for (int qt = qt_lo; qt <= qt_hi; ++qt) {
     __syncthreads();
     stage_R<D>(Qb + (long long)(qt * KT) * qs, qs, q_row);
-    stage_R<D>(dOb + (long long)(qt * KT) * qs, qs, do_row);
+    stage_R<D>(dOb + (long long)(qt * KT) * do_rs, do_rs, do_row);
     stage_T<D>(Qb + (long long)(qt * KT) * qs, qs, qT_lds);
-    stage_T<D>(dOb + (long long)(qt * KT) * qs, qs, doT_lds);
+    stage_T<D>(dOb + (long long)(qt * KT) * do_rs, do_rs, doT_lds);
     __syncthreads();
     // tile fully before this wave's kv rows → all masked: skip compute
     if (qt * KT + KT - 1 < kv_wave_min) continue;
@@ -322,15 +326,15 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
   }
 
-  u16* dKp = dk + ((long long)b * S + kv0) * qs + (long long)h * D;
-  u16* dVp = dv + ((long long)b * S + kv0) * qs + (long long)h * D;
+  u16* dKp = dk + ((long long)b * S + kv0) * od + (long long)h * D;
+  u16* dVp = dv + ((long long)b * S + kv0) * od + (long long)h * D;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
 #pragma unroll
     for (int t = 0; t < DT; ++t) {
-      dKp[(long long)krow * qs + t * 32 + lq] = f32_to_bf16(acc_dk[t][r]);
-      dVp[(long long)krow * qs + t * 32 + lq] = f32_to_bf16(acc_dv[t][r]);
+      dKp[(long long)krow * od + t * 32 + lq] = f32_to_bf16(acc_dk[t][r]);
+      dVp[(long long)krow * od + t * 32 + lq] = f32_to_bf16(acc_dv[t][r]);
     }
   }
 }
@@ -342,32 +346,35 @@ extern "C" {
 void acco_attn_bwd32_dq(const void* q, const void* k, const void* v,
                         const void* dO, const float* lse, const float* delta,
                         void* dq, int B, int S, int H, int Hkv, int D,
-                        float scale, int window, hipStream_t stream) {
+                        float scale, int window, long long q_rs,
+                        long long kv_rs, long long do_rs, long long dq_rs,
+                        hipStream_t stream) {
   dim3 grid(S / BT, B * H);
   const int lds = (2 * KT * (D + 8) + D * LST) * sizeof(u16);
   if (D == 64)
     hipLaunchKernelGGL(attn_bwd32_dq_kernel<64>, grid, dim3(512), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v,
                        (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
-                       scale, window);
+                       scale, window, q_rs, kv_rs, do_rs, dq_rs);
   else
     hipLaunchKernelGGL(attn_bwd32_dq_kernel<128>, grid, dim3(512), lds,
                        stream, (const u16*)q, (const u16*)k, (const u16*)v,
                        (const u16*)dO, lse, delta, (u16*)dq, S, H, Hkv,
-                       scale, window);
+                       scale, window, q_rs, kv_rs, do_rs, dq_rs);
 }
 
 void acco_attn_bwd32_dkv(const void* q, const void* k, const void* v,
                          const void* dO, const float* lse,
                          const float* delta, void* dk, void* dv, int B,
                          int S, int H, int Hkv, int D, float scale,
-                         int window, hipStream_t stream) {
+                         int window, long long q_rs, long long kv_rs,
+                         long long do_rs, hipStream_t stream) {
   dim3 grid(S / BT, B * H);
   const int lds = (2 * KT * (D + 8) + 2 * D * LST) * sizeof(u16);
   hipLaunchKernelGGL(attn_bwd32_dkv_kernel<64>, grid, dim3(512), lds, stream,
                      (const u16*)q, (const u16*)k, (const u16*)v,
                      (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                     Hkv, scale, window);
+                     Hkv, scale, window, q_rs, kv_rs, do_rs);
 }
 
 }  // extern "C"

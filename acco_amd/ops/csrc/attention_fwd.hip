@@ -241,7 +241,8 @@ void attn_fwd_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 extern "C" void acco_attn_fwd32(const void* q, const void* k, const void* v,
                                 void* o, float* lse, int B, int S, int H,
                                 int Hkv, int D, float scale, int window,
-                                hipStream_t stream);
+                                long long q_rs, long long kv_rs,
+                                long long o_rs, hipStream_t stream);
 
 extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               void* o, float* lse, int B, int S, int H,
@@ -249,7 +250,9 @@ extern "C" void acco_attn_fwd(const void* q, const void* k, const void* v,
                               hipStream_t stream) {
   if ((D == 64 || D == 128) && S % 256 == 0) {
     // v4: 32x32 MFMA + in-register softmax (attention_fwd32.hip)
-    acco_attn_fwd32(q, k, v, o, lse, B, S, H, Hkv, D, scale, window, stream);
+    acco_attn_fwd32(q, k, v, o, lse, B, S, H, Hkv, D, scale, window,
+                    (long long)H * D, (long long)Hkv * D, (long long)H * D,
+                    stream);
     return;
   }
   const int lds_bytes =

@@ -40,7 +40,8 @@ __global__ __launch_bounds__(512)
 void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                        const u16* __restrict__ v, u16* __restrict__ o,
                        float* __restrict__ lse,        // [B, H, S]
-                       int S, int H, int Hkv, float scale, int window) {
+                       int S, int H, int Hkv, float scale, int window,
+                       long long q_rs, long long kv_rs, long long o_rs) {
   constexpr int KS = D / 16;       // QK^T K-steps over head dim (16 each)
   constexpr int DT = D / 32;       // 32-wide d tiles of the output
   const int qt = blockIdx.x;
@@ -58,8 +59,8 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   constexpr int KROW = D + 8;
 
   const int q0 = qt * QT + wave * QW;
-  const long long qs = (long long)H * D;
-  const long long ks = (long long)Hkv * D;
+  const long long qs = q_rs;           // token-row strides (packed-qkv aware)
+  const long long ks = kv_rs;
   const u16* Qp = q + ((long long)b * S + q0) * qs + (long long)h * D;
   const u16* Kb = k + (long long)b * S * ks + (long long)hkv * D;
   const u16* Vb = v + (long long)b * S * ks + (long long)hkv * D;
@@ -224,14 +225,14 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 #pragma unroll
   for (int r = 0; r < 16; ++r)
     l_row[r] = __shfl(l_c, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
-  u16* Op = o + ((long long)b * S + q0) * qs + (long long)h * D;
+  u16* Op = o + ((long long)b * S + q0) * o_rs + (long long)h * D;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const float inv_l = (l_row[r] > 0.f) ? 1.0f / l_row[r] : 0.0f;
     const int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
 #pragma unroll
     for (int t = 0; t < DT; ++t)
-      Op[(long long)qrow * qs + t * 32 + lq] =
+      Op[(long long)qrow * o_rs + t * 32 + lq] =
           f32_to_bf16(acc_o[t][r] * inv_l);
   }
   if (hi == 0)
@@ -244,15 +245,16 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 extern "C" void acco_attn_fwd32(const void* q, const void* k, const void* v,
                                 void* o, float* lse, int B, int S, int H,
                                 int Hkv, int D, float scale, int window,
-                                hipStream_t stream) {
+                                long long q_rs, long long kv_rs,
+                                long long o_rs, hipStream_t stream) {
   dim3 grid(S / QT, B * H);
   const int lds = (KT * (D + 8) + D * LST) * sizeof(u16);
   if (D == 64)
     hipLaunchKernelGGL(attn_fwd32_kernel<64>, grid, dim3(512), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v, (u16*)o,
-                       lse, S, H, Hkv, scale, window);
+                       lse, S, H, Hkv, scale, window, q_rs, kv_rs, o_rs);
   else
     hipLaunchKernelGGL(attn_fwd32_kernel<128>, grid, dim3(512), lds, stream,
                        (const u16*)q, (const u16*)k, (const u16*)v, (u16*)o,
-                       lse, S, H, Hkv, scale, window);
+                       lse, S, H, Hkv, scale, window, q_rs, kv_rs, o_rs);
 }

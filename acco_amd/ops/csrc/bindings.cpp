@@ -31,7 +31,8 @@ extern "C" void acco_layernorm_bwd(const void*, const void*, const void*,
                                    const void*, const void*, void*, void*,
                                    void*, long long, int, hipStream_t);
 extern "C" void acco_rope(const void*, void*, const float*, const float*,
-                          long long, int, int, int, bool, hipStream_t);
+                          long long, int, int, int, bool, long long,
+                          hipStream_t);
 extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
                             long long, int, int, hipStream_t);
 extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
@@ -40,6 +41,19 @@ extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
 extern "C" void acco_attn_fwd(const void*, const void*, const void*, void*,
                               float*, int, int, int, int, int, float, int,
                               hipStream_t);
+extern "C" void acco_attn_fwd32(const void*, const void*, const void*, void*,
+                                float*, int, int, int, int, int, float, int,
+                                long long, long long, long long, hipStream_t);
+extern "C" void acco_attn_bwd32_dq(const void*, const void*, const void*,
+                                   const void*, const float*, const float*,
+                                   void*, int, int, int, int, int, float,
+                                   int, long long, long long, long long,
+                                   long long, hipStream_t);
+extern "C" void acco_attn_bwd32_dkv(const void*, const void*, const void*,
+                                    const void*, const float*, const float*,
+                                    void*, void*, int, int, int, int, int,
+                                    float, int, long long, long long,
+                                    long long, hipStream_t);
 extern "C" void acco_attn_bwd_dq(const void*, const void*, const void*,
                                  const void*, const float*, const float*,
                                  void*, int, int, int, int, int, float, int,
@@ -232,7 +246,8 @@ at::Tensor rope_fwd(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
   TORCH_CHECK(D % 8 == 0 && cos_t.size(0) >= S && cos_t.size(1) == D);
   auto y = at::empty_like(x);
   acco_rope(x.data_ptr(), y.data_ptr(), cos_t.data_ptr<float>(),
-            sin_t.data_ptr<float>(), B, S, H, D, bwd, cur_stream());
+            sin_t.data_ptr<float>(), B, S, H, D, bwd,
+            (long long)H * D, cur_stream());
   return y;
 }
 
@@ -320,6 +335,65 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// ---- packed-QKV attention path (fused projection output consumed and
+// grad produced with ZERO split/cat copies). qkv: [B, S, W] contiguous with
+// W = (H + 2*Hkv)*D, sections in q|k|v order; D=64, S%256==0 (v4 kernels).
+std::vector<at::Tensor> rope_packed(at::Tensor src, at::Tensor dst,
+                                    at::Tensor cos_t, at::Tensor sin_t,
+                                    int64_t off, int64_t Hsec, int64_t D,
+                                    bool bwd) {
+  CHECK_BF16_CONTIG(src); CHECK_BF16_CONTIG(dst);
+  const long long B = src.size(0);
+  const int S = (int)src.size(1);
+  const long long W = src.size(2);
+  acco_rope((const u16*)src.data_ptr() + off,
+            (u16*)dst.data_ptr() + off, cos_t.data_ptr<float>(),
+            sin_t.data_ptr<float>(), B, S, (int)Hsec, (int)D, bwd, W,
+            cur_stream());
+  return {dst};
+}
+
+std::vector<at::Tensor> attn_fwd_packed(at::Tensor qkv, int64_t H,
+                                        int64_t Hkv, int64_t D, double scale,
+                                        int64_t window) {
+  CHECK_BF16_CONTIG(qkv);
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1);
+  const long long W = qkv.size(2);
+  TORCH_CHECK(W == (H + 2 * Hkv) * D && D == 64 && S % 256 == 0);
+  const u16* base = (const u16*)qkv.data_ptr();
+  auto o = at::empty({B, S, H * D}, qkv.options());
+  auto lse = at::empty({B, H, S}, qkv.options().dtype(at::kFloat));
+  acco_attn_fwd32(base, base + H * D, base + (H + Hkv) * D, o.data_ptr(),
+                  lse.data_ptr<float>(), B, S, (int)H, (int)Hkv, (int)D,
+                  (float)scale, (int)window, W, W, H * D, cur_stream());
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd_packed(at::Tensor qkv, at::Tensor dO,
+                                        at::Tensor lse, at::Tensor delta,
+                                        at::Tensor dqkv, int64_t H,
+                                        int64_t Hkv, int64_t D, double scale,
+                                        int64_t window) {
+  CHECK_BF16_CONTIG(qkv); CHECK_BF16_CONTIG(dO); CHECK_BF16_CONTIG(dqkv);
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1);
+  const long long W = qkv.size(2);
+  const u16* base = (const u16*)qkv.data_ptr();
+  u16* dbase = (u16*)dqkv.data_ptr();
+  // dq straight into the packed grad (stride W); dk/dv per-QUERY-head temps
+  auto dkq = at::empty({B, S, H, D}, qkv.options());
+  auto dvq = at::empty({B, S, H, D}, qkv.options());
+  acco_attn_bwd32_dq(base, base + H * D, base + (H + Hkv) * D, dO.data_ptr(),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(), dbase,
+                     B, S, (int)H, (int)Hkv, (int)D, (float)scale,
+                     (int)window, W, W, H * D, W, cur_stream());
+  acco_attn_bwd32_dkv(base, base + H * D, base + (H + Hkv) * D,
+                      dO.data_ptr(), lse.data_ptr<float>(),
+                      delta.data_ptr<float>(), dkq.data_ptr(),
+                      dvq.data_ptr(), B, S, (int)H, (int)Hkv, (int)D,
+                      (float)scale, (int)window, W, W, H * D, cur_stream());
+  return {dkq, dvq};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -342,5 +416,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);
+  m.def("rope_packed", &rope_packed);
+  m.def("attn_fwd_packed", &attn_fwd_packed);
+  m.def("attn_bwd_packed", &attn_bwd_packed);
   m.attr("_gfx950") = true;
 }

@@ -17,7 +17,8 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ y,
                             const float* __restrict__ cos_t,
                             const float* __restrict__ sin_t,
                             long long total_q,   // B*S*H*(D/8)
-                            int H, int D, int S) {
+                            int H, int D, int S,
+                            long long row_stride) {  // elements per token row
   const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
   const int qph = D / 8;                 // 4-pair quads per half-head
@@ -25,8 +26,9 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ y,
     const int q = (int)(i % qph);        // which 4-pair group in the head
     long long th = i / qph;              // token*H + h
     const long long tok = th / H;
+    const int h = (int)(th % H);
     const int s = (int)(tok % S);
-    const long long base = th * D + q * 4;          // x1 offset
+    const long long base = tok * row_stride + (long long)h * D + q * 4;
     const long long base2 = base + D / 2;           // x2 offset
     ushort4 x1 = *reinterpret_cast<const ushort4*>(x + base);
     ushort4 x2 = *reinterpret_cast<const ushort4*>(x + base2);
@@ -61,13 +63,16 @@ __global__ void rope_kernel(const u16* __restrict__ x, u16* __restrict__ y,
 
 extern "C" void acco_rope(const void* x, void* y, const float* cos_t,
                           const float* sin_t, long long B, int S, int H,
-                          int D, bool bwd, hipStream_t stream) {
+                          int D, bool bwd, long long row_stride,
+                          hipStream_t stream) {
   const long long total_q = B * (long long)S * H * (D / 8);
   const int grid = elementwise_grid(total_q, 256);
   if (bwd)
     hipLaunchKernelGGL(rope_kernel<true>, dim3(grid), dim3(256), 0, stream,
-                       (const u16*)x, (u16*)y, cos_t, sin_t, total_q, H, D, S);
+                       (const u16*)x, (u16*)y, cos_t, sin_t, total_q, H, D, S,
+                       row_stride);
   else
     hipLaunchKernelGGL(rope_kernel<false>, dim3(grid), dim3(256), 0, stream,
-                       (const u16*)x, (u16*)y, cos_t, sin_t, total_q, H, D, S);
+                       (const u16*)x, (u16*)y, cos_t, sin_t, total_q, H, D, S,
+                       row_stride);
 }
