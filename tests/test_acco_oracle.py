@@ -288,3 +288,18 @@ def test_acco_warmup_path_ws2():
     assert torch.equal(res[0]["params"], res[1]["params"])
     assert torch.isfinite(res[0]["params"]).all()
     assert res[0]["count"] >= TARGET
+
+
+def test_acco_matches_oracle_ws4():
+    """Bucket-major geometry + the two-round algebra at world_size 4."""
+    tmpdir = run_distributed(_worker_acco, 4, args=(0,), timeout=300)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(4)]
+    for r in range(1, 4):
+        assert torch.equal(res[0]["params"], res[r]["params"])
+    torch.manual_seed(7)
+    w0 = nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone()
+    traces = {r: res[r]["trace"] for r in range(4)}
+    P, count_tot = oracle_replay(w0, traces, 4, 0)
+    assert count_tot == res[0]["count_grad_tot"]
+    assert torch.allclose(P, res[0]["params"], atol=1e-6, rtol=1e-6)
