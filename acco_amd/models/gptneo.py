@@ -39,9 +39,20 @@ class GPTNeoSelfAttention(nn.Module):
         B, S, d = x.shape
         H, hd = self.cfg.num_heads, self.cfg.head_dim
         if self._fused_kvq is not None:
+            from acco_amd import ops as _ops
             from acco_amd.models.fuse import FusedArenaLinearFn
             w, g, splits = self._fused_kvq
             kvq = FusedArenaLinearFn.apply(x, w, g)
+            window = (self.cfg.window_size
+                      if self.attention_type == "local" else 0)
+            if (kvq.is_cuda and hd == 64 and S % 256 == 0
+                    and _ops.have_kernel("attn_fwd_packed")):
+                from acco_amd.ops.autograd import AttnQKVPackedFn
+                # GPT-Neo packing order k|v|q; no RoPE; no 1/sqrt(d) scale
+                offs = (2 * d, 0, d)
+                o = AttnQKVPackedFn.apply(kvq, None, None, H, H, hd, 1.0,
+                                          window, offs)
+                return self.out_proj(o)
             k, v, q = torch.split(kvq, splits, dim=-1)
             q = q.contiguous().view(B, S, H, hd)
             k = k.contiguous().view(B, S, H, hd)

@@ -57,9 +57,12 @@ class LlamaAttention(nn.Module):
                 # packed core: RoPE + flash attention directly on the fused
                 # projection output, packed grad back — zero split/cat
                 from acco_amd.ops.autograd import AttnQKVPackedFn
+                hd = cfg.head_dim
+                offs = (0, cfg.num_heads * hd,
+                        (cfg.num_heads + cfg.num_kv_heads) * hd)
                 o = AttnQKVPackedFn.apply(qkv, cos, sin, cfg.num_heads,
-                                          cfg.num_kv_heads, cfg.head_dim,
-                                          cfg.head_dim ** -0.5)
+                                          cfg.num_kv_heads, hd,
+                                          hd ** -0.5, 0, offs)
                 return self.o_proj(o)
             q, k, v = torch.split(qkv, splits, dim=-1)
             q = q.contiguous().view(B, S, cfg.num_heads, cfg.head_dim)

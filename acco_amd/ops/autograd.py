@@ -133,52 +133,62 @@ class CausalLMLossFn(torch.autograd.Function):
 
 
 class AttnQKVPackedFn(torch.autograd.Function):
-    """The whole Llama attention core over the PACKED fused-QKV projection
-    output [B, S, (H+2Hkv)·D]: RoPE (strided kernel, q/k sections) +
-    flash attention v4 forward; backward emits one packed dqkv — no
-    torch.split forward copies, no cat backward. D=64, S%256==0."""
+    """The whole attention core over the PACKED fused projection output
+    [B, S, (H+2Hkv)·D]: optional RoPE (strided kernel on the q/k sections) +
+    flash attention v4 forward; backward emits one packed grad — no
+    torch.split forward copies, no cat backward. D=64, S%256==0.
+    Section offsets support both packing orders (Llama q|k|v,
+    GPT-Neo k|v|q); window>0 = GPT-Neo local attention."""
 
     @staticmethod
-    def forward(ctx, qkv, cos, sin, H, Hkv, D, scale):
+    def forward(ctx, qkv, cos, sin, H, Hkv, D, scale, window, offs):
         ext = ops.hip_ext()
         qkv = qkv.contiguous()
-        cos = cos.float().contiguous()
-        sin = sin.float().contiguous()
-        roped = torch.empty_like(qkv)
-        ext.rope_packed(qkv, roped, cos, sin, 0, H, D, False)
-        ext.rope_packed(qkv, roped, cos, sin, H * D, Hkv, D, False)
-        v_off = (H + Hkv) * D
-        roped[..., v_off:].copy_(qkv[..., v_off:])
-        o, lse = ext.attn_fwd_packed(roped, H, Hkv, D, float(scale), 0)
-        ctx.save_for_backward(roped, o, lse, cos, sin)
-        ctx.meta = (H, Hkv, D, float(scale))
+        q_off, k_off, v_off = offs
+        if cos is not None:
+            cos = cos.float().contiguous()
+            sin = sin.float().contiguous()
+            roped = torch.empty_like(qkv)
+            ext.rope_packed(qkv, roped, cos, sin, q_off, H, D, False)
+            ext.rope_packed(qkv, roped, cos, sin, k_off, Hkv, D, False)
+            roped[..., v_off:v_off + Hkv * D].copy_(
+                qkv[..., v_off:v_off + Hkv * D])
+        else:
+            roped = qkv
+        o, lse = ext.attn_fwd_packed(roped, H, Hkv, D, float(scale),
+                                     int(window or 0), q_off, k_off, v_off)
+        ctx.save_for_backward(roped, o, lse,
+                              cos if cos is not None else torch.empty(0),
+                              sin if sin is not None else torch.empty(0))
+        ctx.meta = (H, Hkv, D, float(scale), int(window or 0), offs)
         return o                     # [B, S, H*D]
 
     @staticmethod
     def backward(ctx, dO):
         ext = ops.hip_ext()
         roped, o, lse, cos, sin = ctx.saved_tensors
-        H, Hkv, D, scale = ctx.meta
+        H, Hkv, D, scale, window, offs = ctx.meta
+        q_off, k_off, v_off = offs
         B, S, W = roped.shape
         dO = dO.contiguous()
         delta = ext.attn_delta(dO.view(B, S, H, D), o.view(B, S, H, D))
         dqkv = torch.empty_like(roped)
         dkq, dvq = ext.attn_bwd_packed(roped, dO, lse, delta, dqkv, H, Hkv,
-                                       D, scale, 0)
+                                       D, scale, window, q_off, k_off, v_off)
         rep = H // Hkv
-        k_off, v_off = H * D, (H + Hkv) * D
         if rep > 1:
             dk = dkq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
             dv = dvq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
-            dqkv[..., k_off:v_off] = dk.reshape(B, S, Hkv * D).bfloat16()
-            dqkv[..., v_off:] = dv.reshape(B, S, Hkv * D).bfloat16()
+            dqkv[..., k_off:k_off + Hkv * D] = dk.reshape(B, S, Hkv * D).bfloat16()
+            dqkv[..., v_off:v_off + Hkv * D] = dv.reshape(B, S, Hkv * D).bfloat16()
         else:
-            dqkv[..., k_off:v_off] = dkq.reshape(B, S, Hkv * D)
-            dqkv[..., v_off:] = dvq.reshape(B, S, Hkv * D)
-        # inverse rotation in place on the q/k grad sections
-        ext.rope_packed(dqkv, dqkv, cos, sin, 0, H, D, True)
-        ext.rope_packed(dqkv, dqkv, cos, sin, k_off, Hkv, D, True)
-        return dqkv, None, None, None, None, None, None
+            dqkv[..., k_off:k_off + Hkv * D] = dkq.reshape(B, S, Hkv * D)
+            dqkv[..., v_off:v_off + Hkv * D] = dvq.reshape(B, S, Hkv * D)
+        if cos.numel():
+            # inverse rotation in place on the q/k grad sections
+            ext.rope_packed(dqkv, dqkv, cos, sin, q_off, H, D, True)
+            ext.rope_packed(dqkv, dqkv, cos, sin, k_off, Hkv, D, True)
+        return (dqkv, None, None, None, None, None, None, None, None)
 
 
 class AttentionFn(torch.autograd.Function):

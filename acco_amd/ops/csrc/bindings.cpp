@@ -355,15 +355,16 @@ std::vector<at::Tensor> rope_packed(at::Tensor src, at::Tensor dst,
 
 std::vector<at::Tensor> attn_fwd_packed(at::Tensor qkv, int64_t H,
                                         int64_t Hkv, int64_t D, double scale,
-                                        int64_t window) {
+                                        int64_t window, int64_t q_off,
+                                        int64_t k_off, int64_t v_off) {
   CHECK_BF16_CONTIG(qkv);
   const int B = (int)qkv.size(0), S = (int)qkv.size(1);
   const long long W = qkv.size(2);
-  TORCH_CHECK(W == (H + 2 * Hkv) * D && D == 64 && S % 256 == 0);
+  TORCH_CHECK(W >= (H + 2 * Hkv) * D && D == 64 && S % 256 == 0);
   const u16* base = (const u16*)qkv.data_ptr();
   auto o = at::empty({B, S, H * D}, qkv.options());
   auto lse = at::empty({B, H, S}, qkv.options().dtype(at::kFloat));
-  acco_attn_fwd32(base, base + H * D, base + (H + Hkv) * D, o.data_ptr(),
+  acco_attn_fwd32(base + q_off, base + k_off, base + v_off, o.data_ptr(),
                   lse.data_ptr<float>(), B, S, (int)H, (int)Hkv, (int)D,
                   (float)scale, (int)window, W, W, H * D, cur_stream());
   return {o, lse};
@@ -373,7 +374,8 @@ std::vector<at::Tensor> attn_bwd_packed(at::Tensor qkv, at::Tensor dO,
                                         at::Tensor lse, at::Tensor delta,
                                         at::Tensor dqkv, int64_t H,
                                         int64_t Hkv, int64_t D, double scale,
-                                        int64_t window) {
+                                        int64_t window, int64_t q_off,
+                                        int64_t k_off, int64_t v_off) {
   CHECK_BF16_CONTIG(qkv); CHECK_BF16_CONTIG(dO); CHECK_BF16_CONTIG(dqkv);
   const int B = (int)qkv.size(0), S = (int)qkv.size(1);
   const long long W = qkv.size(2);
@@ -382,11 +384,12 @@ std::vector<at::Tensor> attn_bwd_packed(at::Tensor qkv, at::Tensor dO,
   // dq straight into the packed grad (stride W); dk/dv per-QUERY-head temps
   auto dkq = at::empty({B, S, H, D}, qkv.options());
   auto dvq = at::empty({B, S, H, D}, qkv.options());
-  acco_attn_bwd32_dq(base, base + H * D, base + (H + Hkv) * D, dO.data_ptr(),
-                     lse.data_ptr<float>(), delta.data_ptr<float>(), dbase,
-                     B, S, (int)H, (int)Hkv, (int)D, (float)scale,
-                     (int)window, W, W, H * D, W, cur_stream());
-  acco_attn_bwd32_dkv(base, base + H * D, base + (H + Hkv) * D,
+  acco_attn_bwd32_dq(base + q_off, base + k_off, base + v_off, dO.data_ptr(),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     dbase + q_off, B, S, (int)H, (int)Hkv, (int)D,
+                     (float)scale, (int)window, W, W, H * D, W,
+                     cur_stream());
+  acco_attn_bwd32_dkv(base + q_off, base + k_off, base + v_off,
                       dO.data_ptr(), lse.data_ptr<float>(),
                       delta.data_ptr<float>(), dkq.data_ptr(),
                       dvq.data_ptr(), B, S, (int)H, (int)Hkv, (int)D,
