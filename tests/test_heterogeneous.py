@@ -58,8 +58,10 @@ def test_acco_heterogeneous_accumulation_ws2():
                       weights_only=False) for r in range(2)]
     assert torch.equal(res[0]["params"], res[1]["params"])
     assert res[0]["count"] >= TARGET
-    # rank 1 contributed ~3x the grads per round
-    assert sum(res[1]["trace"]) > sum(res[0]["trace"])
+    # (no ordering assertion between the ranks' totals: on a loaded CPU the
+    # n_acc=1 rank can legally over-accumulate while coms run — exactly the
+    # heterogeneity the weighted averaging absorbs; the oracle replay below
+    # is the correctness check for ANY recorded schedule)
 
     torch.manual_seed(7)
     w0 = nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone()
