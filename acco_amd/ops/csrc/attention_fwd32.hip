@@ -87,52 +87,36 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const int j_hi = (qt * QT + QT - 1) / KT;
   const int q_wave_max = q0 + QW - 1;
 
-  // T14 async staging: the NEXT tile's K/V global loads are issued right
-  // after this tile's LDS writes and stay in flight under the MFMA work
-  // (plain loads survive __syncthreads; hipcc inserts the counted vmcnt
-  // before the next iteration's ds_writes). Per-thread staging registers:
-  // K D/64 × uint4, V 4 × ushort4.
-  constexpr int KCH = D / 64;            // K uint4 chunks per thread
-  uint4 kreg[KCH];
-  ushort4 va0, va1, vb0, vb1;
-  const int kv2 = (threadIdx.x & 31) * 2;
-  const int vdg = threadIdx.x >> 5;      // < 16 = D/8 max
-  auto load_tile = [&](int j) {
-    const u16* Kt = Kb + (long long)(j * KT) * ks;
-#pragma unroll
-    for (int c = 0; c < KCH; ++c) {
-      const int cc = threadIdx.x + c * 512;
-      const int kv = cc / (D / 8), dc = cc % (D / 8);
-      kreg[c] = *reinterpret_cast<const uint4*>(
-          Kt + (long long)kv * ks + dc * 8);
-    }
-    const u16* Vt = Vb + (long long)(j * KT) * ks;
-    va0 = reinterpret_cast<const ushort4*>(Vt + (long long)kv2 * ks + vdg * 8)[0];
-    va1 = reinterpret_cast<const ushort4*>(Vt + (long long)kv2 * ks + vdg * 8)[1];
-    vb0 = reinterpret_cast<const ushort4*>(Vt + (long long)(kv2 + 1) * ks + vdg * 8)[0];
-    vb1 = reinterpret_cast<const ushort4*>(Vt + (long long)(kv2 + 1) * ks + vdg * 8)[1];
-  };
-  auto write_tile = [&]() {
-#pragma unroll
-    for (int c = 0; c < KCH; ++c) {
-      const int cc = threadIdx.x + c * 512;
-      const int kv = cc / (D / 8), dc = cc % (D / 8);
-      reinterpret_cast<uint4*>(k_lds + kv * KROW)[dc] = kreg[c];
-    }
-    u16 av[8] = {va0.x, va0.y, va0.z, va0.w, va1.x, va1.y, va1.z, va1.w};
-    u16 bv[8] = {vb0.x, vb0.y, vb0.z, vb0.w, vb1.x, vb1.y, vb1.z, vb1.w};
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-      *reinterpret_cast<ushort2*>(v_lds + (vdg * 8 + i) * LST + kv2) =
-          make_ushort2(av[i], bv[i]);
-  };
-
-  load_tile(j_lo);
   for (int j = j_lo; j <= j_hi; ++j) {
-    __syncthreads();                     // prior tile's LDS reads complete
-    write_tile();
+    // ---- stage K (row-major copy) + V (transposed) for all 8 waves
     __syncthreads();
-    if (j < j_hi) load_tile(j + 1);      // in flight under the MFMAs below
+    {
+      const u16* Kt = Kb + (long long)(j * KT) * ks;
+      for (int c = threadIdx.x; c < KT * (D / 8); c += 512) {
+        const int kv = c / (D / 8), dc = c % (D / 8);
+        reinterpret_cast<uint4*>(k_lds + kv * KROW)[dc] =
+            *reinterpret_cast<const uint4*>(Kt + (long long)kv * ks + dc * 8);
+      }
+      const u16* Vt = Vb + (long long)(j * KT) * ks;
+      const int kv2 = (threadIdx.x & 31) * 2;
+      for (int dg = threadIdx.x >> 5; dg < D / 8; dg += 16) {
+        ushort4 a0 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)kv2 * ks + dg * 8)[0];
+        ushort4 a1 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)kv2 * ks + dg * 8)[1];
+        ushort4 b0 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)(kv2 + 1) * ks + dg * 8)[0];
+        ushort4 b1 = reinterpret_cast<const ushort4*>(
+            Vt + (long long)(kv2 + 1) * ks + dg * 8)[1];
+        u16 av[8] = {a0.x, a0.y, a0.z, a0.w, a1.x, a1.y, a1.z, a1.w};
+        u16 bv[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          *reinterpret_cast<ushort2*>(v_lds + (dg * 8 + i) * LST + kv2) =
+              make_ushort2(av[i], bv[i]);
+      }
+    }
+    __syncthreads();
 
     // fully-masked tile for this wave (kv all in the future): skip compute
     if (j * KT > q_wave_max) continue;
