@@ -38,6 +38,8 @@ extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
 extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
                             void*, const float*, float, long long, int, int,
                             hipStream_t);
+extern "C" void acco_gemm_nt(const void*, const void*, void*, int, int,
+                             int, hipStream_t);
 extern "C" void acco_attn_fwd(const void*, const void*, const void*, void*,
                               float*, int, int, int, int, int, float, int,
                               hipStream_t);
@@ -335,6 +337,17 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// ---- experimental NT GEMM (C = A · B^T, bf16; bench/refcheck only)
+at::Tensor gemm_nt(at::Tensor A, at::Tensor B) {
+  CHECK_BF16_CONTIG(A); CHECK_BF16_CONTIG(B);
+  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
+  TORCH_CHECK(B.size(1) == K && M % 128 == 0 && N % 128 == 0 && K % 64 == 0);
+  auto C = at::empty({M, N}, A.options());
+  acco_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr(), M, N, K,
+               cur_stream());
+  return C;
+}
+
 // ---- packed-QKV attention path (fused projection output consumed and
 // grad produced with ZERO split/cat copies). qkv: [B, S, W] contiguous with
 // W = (H + 2*Hkv)*D, sections in q|k|v order; D=64, S%256==0 (v4 kernels).
@@ -419,6 +432,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);
+  m.def("gemm_nt", &gemm_nt);
   m.def("rope_packed", &rope_packed);
   m.def("attn_fwd_packed", &attn_fwd_packed);
   m.def("attn_bwd_packed", &attn_bwd_packed);
