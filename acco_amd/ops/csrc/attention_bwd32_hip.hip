@@ -259,13 +259,68 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const int qt_lo = (jb * BT) / KT;
   const int kv_wave_min = kv0;
 
+  // T14 async staging: hold the next q tile's Q/dO loads in registers
+  // across the barrier (+16 VGPR — still 2 waves/SIMD); the loads stay in
+  // flight under the MFMA work. Transpose-staging register pieces:
+  const int t_r2 = (threadIdx.x & 31) * 2;
+  const int t_dg = threadIdx.x >> 5;
+  const bool t_act = t_dg < D / 8;       // D=64: upper waves idle in stage_T
+  ushort4 sq[4], sdo[4];
+  // row-major staging piece: one uint4 per thread per tensor (KT*D/8/512)
+  static_assert(KT * (D / 8) % 512 == 0, "rowmajor chunks");
+  constexpr int RCH = KT * (D / 8) / 512;
+  uint4 rq[RCH], rdo[RCH];
+
+  auto load_qtile = [&](int qt) {
+    const u16* Qs = Qb + (long long)(qt * KT) * qs;
+    const u16* Ds = dOb + (long long)(qt * KT) * do_rs;
+    if (t_act) {
+      sq[0] = reinterpret_cast<const ushort4*>(Qs + (long long)t_r2 * qs + t_dg * 8)[0];
+      sq[1] = reinterpret_cast<const ushort4*>(Qs + (long long)t_r2 * qs + t_dg * 8)[1];
+      sq[2] = reinterpret_cast<const ushort4*>(Qs + (long long)(t_r2 + 1) * qs + t_dg * 8)[0];
+      sq[3] = reinterpret_cast<const ushort4*>(Qs + (long long)(t_r2 + 1) * qs + t_dg * 8)[1];
+      sdo[0] = reinterpret_cast<const ushort4*>(Ds + (long long)t_r2 * do_rs + t_dg * 8)[0];
+      sdo[1] = reinterpret_cast<const ushort4*>(Ds + (long long)t_r2 * do_rs + t_dg * 8)[1];
+      sdo[2] = reinterpret_cast<const ushort4*>(Ds + (long long)(t_r2 + 1) * do_rs + t_dg * 8)[0];
+      sdo[3] = reinterpret_cast<const ushort4*>(Ds + (long long)(t_r2 + 1) * do_rs + t_dg * 8)[1];
+    }
+#pragma unroll
+    for (int c = 0; c < RCH; ++c) {
+      const int cc = threadIdx.x + c * 512;
+      const int r = cc / (D / 8), dc = cc % (D / 8);
+      rq[c] = *reinterpret_cast<const uint4*>(Qs + (long long)r * qs + dc * 8);
+      rdo[c] = *reinterpret_cast<const uint4*>(Ds + (long long)r * do_rs + dc * 8);
+    }
+  };
+  auto write_qtile = [&]() {
+    if (t_act) {
+      u16 aq[8] = {sq[0].x, sq[0].y, sq[0].z, sq[0].w, sq[1].x, sq[1].y, sq[1].z, sq[1].w};
+      u16 bq[8] = {sq[2].x, sq[2].y, sq[2].z, sq[2].w, sq[3].x, sq[3].y, sq[3].z, sq[3].w};
+      u16 ad[8] = {sdo[0].x, sdo[0].y, sdo[0].z, sdo[0].w, sdo[1].x, sdo[1].y, sdo[1].z, sdo[1].w};
+      u16 bd[8] = {sdo[2].x, sdo[2].y, sdo[2].z, sdo[2].w, sdo[3].x, sdo[3].y, sdo[3].z, sdo[3].w};
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        *reinterpret_cast<ushort2*>(qT_lds + (t_dg * 8 + i) * LST + t_r2) =
+            make_ushort2(aq[i], bq[i]);
+        *reinterpret_cast<ushort2*>(doT_lds + (t_dg * 8 + i) * LST + t_r2) =
+            make_ushort2(ad[i], bd[i]);
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < RCH; ++c) {
+      const int cc = threadIdx.x + c * 512;
+      const int r = cc / (D / 8), dc = cc % (D / 8);
+      reinterpret_cast<uint4*>(q_row + r * KROW)[dc] = rq[c];
+      reinterpret_cast<uint4*>(do_row + r * KROW)[dc] = rdo[c];
+    }
+  };
+
+  load_qtile(qt_lo);
   for (int qt = qt_lo; qt <= qt_hi; ++qt) {
     __syncthreads();
-    stage_R<D>(Qb + (long long)(qt * KT) * qs, qs, q_row);
-    stage_R<D>(dOb + (long long)(qt * KT) * do_rs, do_rs, do_row);
-    stage_T<D>(Qb + (long long)(qt * KT) * qs, qs, qT_lds);
-    stage_T<D>(dOb + (long long)(qt * KT) * do_rs, do_rs, doT_lds);
+    write_qtile();
     __syncthreads();
+    if (qt < qt_hi) load_qtile(qt + 1);  // in flight under the MFMAs
     // tile fully before this wave's kv rows → all masked: skip compute
     if (qt * KT + KT - 1 < kv_wave_min) continue;
 
