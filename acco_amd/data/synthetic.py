@@ -32,3 +32,24 @@ def collate_input_ids(batch):
     """Stack const-length sequences (reference trainer_base.py:131-132)."""
     return {"input_ids": torch.stack(
         [torch.as_tensor(b["input_ids"], dtype=torch.long) for b in batch])}
+
+
+def make_padded_collator(pad_token_id: int, pad_to_multiple: int = 64):
+    """Ragged-batch collator for the finetune path (const_len_batch=false):
+    right-pads input_ids and masks padded positions with -100 in labels —
+    the behaviour of the reference's DataCollatorForLanguageModeling
+    (trainer_base.py:209). Padding to a multiple of 64 keeps the flash
+    kernels' S%64 contract."""
+    def collate(batch):
+        seqs = [torch.as_tensor(b["input_ids"], dtype=torch.long)
+                for b in batch]
+        longest = max(s.numel() for s in seqs)
+        target = ((longest + pad_to_multiple - 1) // pad_to_multiple
+                  * pad_to_multiple)
+        ids = torch.full((len(seqs), target), pad_token_id, dtype=torch.long)
+        labels = torch.full((len(seqs), target), -100, dtype=torch.long)
+        for i, s in enumerate(seqs):
+            ids[i, :s.numel()] = s
+            labels[i, :s.numel()] = s
+        return {"input_ids": ids, "labels": labels}
+    return collate

@@ -183,9 +183,19 @@ class DecoupledTrainer:
         # torch Datasets (synthetic) are assumed already per-rank seeded
 
     def _make_dataloader(self, dataset, shuffle: bool) -> DataLoader:
+        if getattr(self.args, "const_len_batch", True):
+            collate = collate_input_ids
+        else:
+            # finetune path: ragged sequences, right-padded with labels
+            # masked at pads (reference DataCollatorForLanguageModeling,
+            # trainer_base.py:209)
+            from acco_amd.data.synthetic import make_padded_collator
+            pad_id = (self.tokenizer.eos_token_id
+                      if self.tokenizer is not None else 0)
+            collate = make_padded_collator(pad_id)
         kwargs = dict(
             batch_size=self.args.batch_size,
-            collate_fn=collate_input_ids,
+            collate_fn=collate,
             num_workers=int(self.args.dataloader_num_workers or 0),
             drop_last=True,
         )

@@ -35,3 +35,13 @@ def test_synthetic_dataset_deterministic():
     batch = collate_input_ids([ds[0], ds[1]])
     assert batch["input_ids"].shape == (2, 16)
     assert batch["input_ids"].dtype == torch.long
+
+
+def test_padded_collator():
+    from acco_amd.data.synthetic import make_padded_collator
+    c = make_padded_collator(pad_token_id=9, pad_to_multiple=8)
+    out = c([{"input_ids": [1, 2, 3]}, {"input_ids": [4, 5, 6, 7, 8]}])
+    assert out["input_ids"].shape == (2, 8)
+    assert out["input_ids"][0].tolist() == [1, 2, 3, 9, 9, 9, 9, 9]
+    assert out["labels"][0].tolist() == [1, 2, 3, -100, -100, -100, -100, -100]
+    assert out["labels"][1].tolist() == [4, 5, 6, 7, 8, -100, -100, -100]

@@ -128,3 +128,61 @@ def test_trainer_acco_bf16_ws2():
                       weights_only=False) for r in range(2)]
     assert torch.equal(res[0]["params"], res[1]["params"])
     assert torch.isfinite(res[0]["params"]).all()
+
+
+class _RaggedDS(torch.utils.data.Dataset):
+    def __init__(self, n, vocab, seed):
+        self.n, self.vocab, self.seed = n, vocab, seed
+
+    def __len__(self):
+        return self.n
+
+    def __getitem__(self, i):
+        g = torch.Generator().manual_seed(self.seed + i)
+        L = int(torch.randint(4, 14, (1,), generator=g))
+        return {"input_ids": torch.randint(0, self.vocab, (L,), generator=g)}
+
+
+def _worker_acco_ft(rank, world, port, tmpdir):
+    """Finetune-shaped path: ragged batches, padded collator, label
+    masking (const_len_batch=false)."""
+    os.environ.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                       "RANK": str(rank), "LOCAL_RANK": str(rank),
+                       "WORLD_SIZE": str(world)})
+    os.chdir(tmpdir)
+    from acco_amd.config import load_config
+    from acco_amd.engine.trainer import DecoupledTrainer
+    from acco_amd.models import GPTNeoConfig, GPTNeoForCausalLM
+
+    cfg = load_config([
+        "train=acco-ft", "train.nb_steps_tot=6", "train.batch_size=2",
+        "train.use_mixed_precision=false", "train.save=false",
+        "train.eval=false", "train.n_grad_accumulation=1",
+        "train.dataloader_num_workers=0", "train.comm_buckets=2",
+        "train.dataloader_persistent_workers=false",
+    ])
+    torch.manual_seed(42)
+    mcfg = GPTNeoConfig(hidden_size=32, num_layers=1, num_heads=2,
+                        vocab_size=64, max_position_embeddings=128,
+                        window_size=8)
+    model = GPTNeoForCausalLM(mcfg)
+
+    class Tok:
+        eos_token_id = 0
+
+    trainer = DecoupledTrainer(model=model, tokenizer=Tok(),
+                               train_dataset=_RaggedDS(32, 64, 5 + rank),
+                               eval_dataset=None, args=cfg.train,
+                               run_name="ft")
+    trainer.train()
+    torch.save({"params": trainer.params[:trainer.n_live].clone()},
+               os.path.join(tmpdir, f"p_ft_{rank}.pt"))
+    teardown_worker()
+
+
+def test_trainer_acco_finetune_ragged_ws2():
+    tmpdir = run_distributed(_worker_acco_ft, 2, timeout=300)
+    res = [torch.load(os.path.join(tmpdir, f"p_ft_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert torch.isfinite(res[0]["params"]).all()
