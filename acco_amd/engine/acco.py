@@ -289,7 +289,8 @@ class AccoEngine:
 
     # ---------------------------------------------------------- DPU loop
 
-    def train_dpu(self, nb_grad_tot: int, n_warmup_steps: int = 0) -> None:
+    def train_dpu(self, nb_grad_tot: int, n_warmup_steps: int = 0,
+                  max_rounds: Optional[int] = None) -> None:
         """Delayed-parameter-update baseline: same primitives, sequential,
         stale gradients, full step every round (reference train_dpu
         :605-663 — communication_step there always runs with the default
@@ -297,12 +298,15 @@ class AccoEngine:
         update_buffers_step alternates grad zeroing)."""
         self.bootstrap(n_warmup_steps)
         count_com = 0
-        while self.count_grad_tot < nb_grad_tot:
+        while self.count_grad_tot < nb_grad_tot and (
+                max_rounds is None or count_com < max_rounds):
             for _ in range(self.n_acc):
                 self.gradient_step(self.next_batch())
             self.communication_round(commit=True)
             self.count_grad_tot += int(self.count_grad_this_round.item())
             self.update_buffers_step(zero_grads=(count_com % 2 == 0))
             count_com += 1
+            if self.on_round_boundary is not None:
+                self.on_round_boundary(count_com)
             if self.on_round_complete is not None and self.rank == 0:
                 self.on_round_complete(count_com, self.count_grad_tot)

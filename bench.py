@@ -59,7 +59,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--method", choices=["acco", "ddp"], default="acco")
+    p.add_argument("--method", choices=["acco", "ddp", "dpu"], default="acco")
     p.add_argument("--model", default="llama-1b")
     p.add_argument("--batch", type=int, default=8)
     p.add_argument("--seq", type=int, default=1024)
@@ -166,7 +166,7 @@ def main():
     steps = args.steps
     state = {"t0": None, "t1": None, "mb0": 0, "mb1": 0}
 
-    if args.method == "acco":
+    if args.method in ("acco", "dpu"):
         eng = AccoEngine(params_arena=params, grads_arena=grads,
                          n_live=n_live, spec=spec, comm=comm, rank=rank,
                          device=device, opt=opt, sched=sched,
@@ -182,8 +182,12 @@ def main():
                 state["t1"] = time.time(); state["mb1"] = eng.micro_steps
 
         eng.on_round_boundary = boundary
-        eng.train_acco(nb_grad_tot=1 << 60, n_warmup_steps=0,
-                       max_rounds=warmup + steps)
+        if args.method == "acco":
+            eng.train_acco(nb_grad_tot=1 << 60, n_warmup_steps=0,
+                           max_rounds=warmup + steps)
+        else:
+            eng.train_dpu(nb_grad_tot=1 << 60, n_warmup_steps=0,
+                          max_rounds=warmup + steps)
         micro = state["mb1"] - state["mb0"]
     else:
         ddp = NativeZeroDDP(model, params, grads, n_live, spec, comm, rank, opt)
@@ -243,8 +247,8 @@ def main():
                 "method": args.method,
                 "global_batch": args.batch * world,
                 "seq_len": args.seq,
-                "parallelism": f"acco-dp{world}" if args.method == "acco"
-                               else f"ddp{world}",
+                "parallelism": (f"{args.method}-dp{world}"
+                                if args.method != "ddp" else f"ddp{world}"),
                 "comm_buckets": spec.nb,
                 "params": n_live,
             },
