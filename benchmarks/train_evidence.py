@@ -29,6 +29,7 @@ def main():
     from acco_amd.models import GPTNeoConfig, GPTNeoForCausalLM
 
     steps = int(sys.argv[1]) if len(sys.argv) > 1 else 200
+    use_llama = "--llama" in sys.argv
     cfg = load_config([
         "train=acco", f"train.nb_steps_tot={steps}", "train.batch_size=8",
         "train.max_length=256", "train.save=false", "train.warmup=20",
@@ -36,10 +37,17 @@ def main():
         "train.dataloader_persistent_workers=false",
     ])
     torch.manual_seed(42)
-    mcfg = GPTNeoConfig(hidden_size=768, num_layers=12, num_heads=12,
-                        vocab_size=50304, max_position_embeddings=2048,
-                        window_size=256)
-    model = GPTNeoForCausalLM(mcfg)
+    if use_llama:
+        from acco_amd.models import LlamaConfig, LlamaForCausalLM
+        mcfg = LlamaConfig(hidden_size=2048, num_layers=16, num_heads=32,
+                           num_kv_heads=8, intermediate_size=8192,
+                           vocab_size=50304, max_position_embeddings=4096)
+        model = LlamaForCausalLM(mcfg)
+    else:
+        mcfg = GPTNeoConfig(hidden_size=768, num_layers=12, num_heads=12,
+                            vocab_size=50304, max_position_embeddings=2048,
+                            window_size=256)
+        model = GPTNeoForCausalLM(mcfg)
     ds = SuccessorDataset(2048, 256, 50304, seed=7)
 
     losses = []
