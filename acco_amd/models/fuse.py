@@ -16,10 +16,24 @@ This is an MI355X-first arena dividend: bigger GEMMs fill the 256-CU chip
 
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import torch
 import torch.nn as nn
+
+# wgrad accumulation flavor: "1" fuses the += into the hipBLASLt epilogue
+# (beta=1); default "0" keeps GEMM-to-temp + vectorized add, measured faster
+# on MI355X (hipBLASLt's beta=1 wgrad solutions run ~5-8% slower than
+# beta=0, outweighing the saved add kernel — profiles/ r01 v6 vs v7).
+_EPILOGUE = os.environ.get("ACCO_WGRAD_EPILOGUE", "0") == "1"
+
+
+def _acc_wgrad(g_view, d2t, x2):
+    if _EPILOGUE:
+        g_view.addmm_(d2t, x2)
+    else:
+        g_view.add_(torch.matmul(d2t, x2))
 
 
 class FusedArenaLinearFn(torch.autograd.Function):
@@ -37,9 +51,41 @@ class FusedArenaLinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         d2 = dout.reshape(-1, dout.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
-        ctx.g_view.add_(torch.matmul(d2.t(), x2))
+        _acc_wgrad(ctx.g_view, d2.t(), x2)
         dx = torch.matmul(dout, w)
         return dx, None, None
+
+
+class ArenaLinearFn(torch.autograd.Function):
+    """Single linear over an arena weight view (optional bias): dW (and db)
+    accumulate in-place into the aliased grad-arena views, bypassing
+    autograd's temp-dW + AccumulateGrad add."""
+
+    @staticmethod
+    def forward(ctx, x, w_view, bias, g_w, g_b):
+        ctx.save_for_backward(x, w_view)
+        ctx.g_w, ctx.g_b = g_w, g_b
+        return torch.nn.functional.linear(x, w_view, bias)
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, w = ctx.saved_tensors
+        d2 = dout.reshape(-1, dout.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        _acc_wgrad(ctx.g_w, d2.t(), x2)
+        if ctx.g_b is not None:
+            ctx.g_b.add_(d2.sum(0))
+        return torch.matmul(dout, w), None, None, None, None
+
+
+def arena_linear(lin: nn.Linear, x: torch.Tensor) -> torch.Tensor:
+    """Run `lin` with in-place grad-arena accumulation if
+    `install_fused_projections` wrapped it; plain nn.Linear otherwise."""
+    fuse = getattr(lin, "_arena_fuse", None)
+    if fuse is None:
+        return lin(x)
+    w_view, g_w, b, g_b = fuse
+    return ArenaLinearFn.apply(x, w_view, b, g_w, g_b)
 
 
 def _arena_views(params: List[nn.Parameter], arena: torch.Tensor,
@@ -73,26 +119,60 @@ def _arena_views(params: List[nn.Parameter], arena: torch.Tensor,
     return w_view, g_view, splits
 
 
+def _param_views(p: nn.Parameter, arena: torch.Tensor,
+                 grads: torch.Tensor) -> Optional[tuple]:
+    """(weight view, grad view) of `arena`/`grads` aliasing param `p`."""
+    if p is None or p.data.untyped_storage().data_ptr() != arena.untyped_storage().data_ptr():
+        return None
+    off, n = p.data.storage_offset(), p.numel()
+    return arena[off:off + n].view(p.shape), grads[off:off + n].view(p.shape)
+
+
+def _wrap_single(lin: nn.Linear, arena: torch.Tensor,
+                 grads: torch.Tensor) -> bool:
+    """Give a plain nn.Linear the in-place-dW arena path (`arena_linear`)."""
+    got = _param_views(lin.weight, arena, grads)
+    if got is None:
+        return False
+    w_view, g_w = got
+    b = g_b = None
+    if lin.bias is not None:
+        gotb = _param_views(lin.bias, arena, grads)
+        if gotb is None:
+            return False
+        b, g_b = gotb
+    lin._arena_fuse = (w_view, g_w, b, g_b)
+    return True
+
+
 def install_fused_projections(model: nn.Module, params_arena: torch.Tensor,
                               grads_arena: torch.Tensor) -> int:
     """Attach fused views to every attention / MLP block whose projections
-    are arena-adjacent. Returns the number of fused groups installed."""
+    are arena-adjacent, and the in-place-dW path to every other arena-resident
+    projection (o/down/out/c_fc/c_proj/lm_head). Returns the number of fused
+    groups installed."""
     from acco_amd.models.gptneo import GPTNeoSelfAttention
     from acco_amd.models.llama import LlamaAttention, LlamaMLP
 
     count = 0
+    grouped = set()
     for mod in model.modules():
         if isinstance(mod, LlamaAttention):
             got = _arena_views([mod.q_proj.weight, mod.k_proj.weight,
                                 mod.v_proj.weight], params_arena, grads_arena)
             if got:
                 mod._fused_qkv = got
+                grouped.update(id(p) for p in (mod.q_proj.weight,
+                                               mod.k_proj.weight,
+                                               mod.v_proj.weight))
                 count += 1
         elif isinstance(mod, LlamaMLP):
             got = _arena_views([mod.gate_proj.weight, mod.up_proj.weight],
                                params_arena, grads_arena)
             if got:
                 mod._fused_gate_up = got
+                grouped.update(id(p) for p in (mod.gate_proj.weight,
+                                               mod.up_proj.weight))
                 count += 1
         elif isinstance(mod, GPTNeoSelfAttention):
             # GPT-Neo declares k, v, q in that order (HF layout)
@@ -100,5 +180,12 @@ def install_fused_projections(model: nn.Module, params_arena: torch.Tensor,
                                 mod.q_proj.weight], params_arena, grads_arena)
             if got:
                 mod._fused_kvq = got
+                grouped.update(id(p) for p in (mod.k_proj.weight,
+                                               mod.v_proj.weight,
+                                               mod.q_proj.weight))
                 count += 1
+    for mod in model.modules():
+        if (isinstance(mod, nn.Linear) and id(mod.weight) not in grouped
+                and mod.weight.dim() == 2):
+            _wrap_single(mod, params_arena, grads_arena)
     return count

@@ -20,6 +20,7 @@ import torch.nn as nn
 
 from acco_amd import ops
 from acco_amd.models.config import LlamaConfig
+from acco_amd.models.fuse import arena_linear
 
 
 class LlamaRMSNorm(nn.Module):
@@ -63,7 +64,7 @@ class LlamaAttention(nn.Module):
                 o = AttnQKVPackedFn.apply(qkv, cos, sin, cfg.num_heads,
                                           cfg.num_kv_heads, hd,
                                           hd ** -0.5, 0, offs)
-                return self.o_proj(o)
+                return arena_linear(self.o_proj, o)
             q, k, v = torch.split(qkv, splits, dim=-1)
             q = q.contiguous().view(B, S, cfg.num_heads, cfg.head_dim)
             k = k.contiguous().view(B, S, cfg.num_kv_heads, cfg.head_dim)
@@ -74,7 +75,7 @@ class LlamaAttention(nn.Module):
             v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
         q, k = ops.rope_apply(q, k, cos, sin)
         o = ops.causal_attention(q, k, v)          # [B, S, H, D]
-        return self.o_proj(o.reshape(B, S, -1))
+        return arena_linear(self.o_proj, o.reshape(B, S, -1))
 
 
 class LlamaMLP(nn.Module):
@@ -92,11 +93,11 @@ class LlamaMLP(nn.Module):
             gu = FusedArenaLinearFn.apply(x, w, g)
             if gu.is_cuda and ops.have_kernel("swiglu_packed_fwd"):
                 from acco_amd.ops.autograd import SwiGLUPackedFn
-                return self.down_proj(SwiGLUPackedFn.apply(gu))
+                return arena_linear(self.down_proj, SwiGLUPackedFn.apply(gu))
             gate, up = torch.split(gu, splits, dim=-1)
-            return self.down_proj(ops.swiglu(gate.contiguous(),
+            return arena_linear(self.down_proj, ops.swiglu(gate.contiguous(),
                                              up.contiguous()))
-        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+        return arena_linear(self.down_proj, ops.swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -170,7 +171,7 @@ class LlamaForCausalLM(nn.Module):
         (logits,) otherwise — the trainer indexes outputs[0]
         (reference trainer_decoupled.py:29-34)."""
         hidden = self.model(input_ids)
-        logits = self.lm_head(hidden)
+        logits = arena_linear(self.lm_head, hidden)
         if labels is None:
             return (logits,)
         loss = ops.causal_lm_loss(logits, labels)
