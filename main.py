@@ -53,6 +53,19 @@ def main(argv=None):
         torch.cuda.manual_seed_all(42)  # reference main.py:28
 
     model = build_model(cfg.model)
+    if cfg.train.get("finetune"):
+        # reference main.py:33-41: finetune starts from pretrained weights.
+        # No-network environment: the checkpoint must be local —
+        # model.pretrained_path (config) or train.pretrained_path override.
+        path = (cfg.model.get("pretrained_path")
+                or cfg.train.get("pretrained_path"))
+        if path:
+            from acco_amd.models import load_pretrained
+            load_pretrained(model, path)
+            logger.info("loaded pretrained weights from %s", path)
+        else:
+            logger.warning("finetune=true but no pretrained_path set; "
+                           "starting from random init")
     logger.info("model instantiated (%.1fM params)",
                 sum(p.numel() for p in model.parameters()) / 1e6)
     train_ds, eval_ds, tokenizer = build_datasets(cfg, model)

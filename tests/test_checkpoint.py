@@ -70,3 +70,32 @@ def test_consolidate_ws2():
     from tests.conftest import run_distributed
     tmpdir = run_distributed(_worker_consolidate, 2, timeout=120)
     assert os.path.exists(os.path.join(tmpdir, "ok.pt"))
+
+
+def test_load_pretrained_file_and_dir(tmp_path):
+    """Finetune entry (reference main.py:33-41 capability): load weights
+    from a state-dict file and from an HF-style model directory."""
+    import torch
+
+    from acco_amd.models import (LlamaConfig, LlamaForCausalLM,
+                                 load_pretrained)
+
+    cfg = LlamaConfig(hidden_size=32, num_layers=2, num_heads=4,
+                      num_kv_heads=2, intermediate_size=64, vocab_size=64,
+                      max_position_embeddings=64)
+    torch.manual_seed(0)
+    src = LlamaForCausalLM(cfg)
+
+    f = tmp_path / "model.pt"
+    torch.save(src.state_dict(), f)
+    m1 = LlamaForCausalLM(cfg)
+    load_pretrained(m1, str(f))
+    for a, b in zip(src.parameters(), m1.parameters()):
+        assert torch.equal(a, b)
+
+    d = tmp_path / "hf_dir"
+    d.mkdir()
+    torch.save(src.state_dict(), d / "pytorch_model.bin")
+    m2 = LlamaForCausalLM(cfg)
+    load_pretrained(m2, str(d))
+    assert torch.equal(m2.lm_head.weight, src.lm_head.weight)
