@@ -9,18 +9,28 @@ namespace {
 
 ACCO_DEV float sigmoidf_(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
+// hardware-exp tanh: one v_exp_f32 + a few VALU ops instead of libm's
+// polynomial tanhf (the kernels are VALU-bound at 99% busy with tanhf —
+// profiles/r01_kernels_pmc.txt). exp overflow at large |x| gives
+// 2/inf = 0 → ±1 exactly; bf16-accurate.
+ACCO_DEV float tanh_fast(float x) {
+  float a = __builtin_fabsf(x);
+  float t = 1.0f - 2.0f / (__expf(2.0f * a) + 1.0f);
+  return __builtin_copysignf(t, x);
+}
+
 // gelu_new: 0.5x(1+tanh(sqrt(2/pi)(x+0.044715x^3)))
 ACCO_DEV float gelu_new_f(float x) {
   const float c = 0.7978845608028654f;    // sqrt(2/pi)
   float u = c * (x + 0.044715f * x * x * x);
-  return 0.5f * x * (1.0f + tanhf(u));
+  return 0.5f * x * (1.0f + tanh_fast(u));
 }
 
 ACCO_DEV float gelu_new_grad_f(float x) {
   const float c = 0.7978845608028654f;
   float x2 = x * x;
   float u = c * (x + 0.044715f * x * x2);
-  float t = tanhf(u);
+  float t = tanh_fast(u);
   float sech2 = 1.0f - t * t;
   float du = c * (1.0f + 3.0f * 0.044715f * x2);
   return 0.5f * (1.0f + t) + 0.5f * x * sech2 * du;
