@@ -23,7 +23,7 @@ extern "C" void acco_rmsnorm_fwd(const void*, const void*, void*, void*,
 extern "C" void acco_rmsnorm_bwd(const void*, const void*, const void*,
                                  const void*, void*, void*, long long, int,
                                  hipStream_t);
-extern "C" int acco_norm_bwd_grid(long long R);
+extern "C" int acco_norm_bwd_grid(long long R, int D);
 extern "C" void acco_layernorm_fwd(const void*, const void*, const void*,
                                    void*, void*, void*, long long, int, float,
                                    hipStream_t);
@@ -195,7 +195,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
   auto dx = at::empty_like(x);
-  const int grid = acco_norm_bwd_grid(R);
+  const int grid = acco_norm_bwd_grid(R, D);
   auto dw_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                    rstd.data_ptr(), dx.data_ptr(), dw_part.data_ptr(), R, D,
@@ -226,7 +226,7 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
   auto dx = at::empty_like(x);
-  const int grid = acco_norm_bwd_grid(R);
+  const int grid = acco_norm_bwd_grid(R, D);
   auto dw_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   auto db_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),

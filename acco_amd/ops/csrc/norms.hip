@@ -1,14 +1,18 @@
 // RMSNorm (Llama) and LayerNorm (GPT-Neo) forward + backward for gfx950.
-// One 256-thread block (4 waves) per row, grid-striding rows; bf16 loads
-// vectorized ×8 (guide G13); row statistics by wave shuffle + LDS
-// cross-wave reduce; weight/bias staged in LDS once per block. dW/dB
-// accumulate per-block partials in registers and land with one fp32
-// atomicAdd per column per block (guide G12).
+// One 256-thread block per G rows (G = 4/2/1 picked from D so every wave
+// has work even at small D — D=768 fills 96 of 256 lanes in the one-row
+// layout, measured 1.4 TB/s; the grouped layout packs 2 rows per block),
+// grid-striding rows; bf16 loads vectorized ×8 (guide G13); row statistics
+// by wave shuffle + LDS cross-wave reduce; weight/bias staged in LDS once
+// per block.
 //
 // The per-thread value arrays are templated on the compile-time chunk
 // count (guide §5.4 rule 20: runtime-indexed ext-vector arrays go to
 // scratch — a first version with a runtime chunk loop ran 8-14× off
 // roofline; see profiles/r01_bench_llama1b_acco_1gpu_kernels.txt).
+// dW/dB land as one [partial_rows, D] fp32 scratch row per (block, group)
+// reduced by a tiny torch sum in the wrapper (atomicAdd serialized ~3×
+// worse).
 // Replaces the HF RMSNorm / nn.LayerNorm ATen chains.
 
 #include "common.h"
@@ -20,14 +24,19 @@ using u16 = unsigned short;
 constexpr int BLOCK = 256;
 constexpr int VEC = 8;                  // bf16 per thread per chunk
 
-// block-reduce a single float (sum) over 4 waves
-ACCO_DEV float block_reduce_sum(float x, float* lds) {
+// Reduce a float over one row-group (TPG = BLOCK/G threads). All groups
+// hit the same __syncthreads unconditionally — barriers stay block-uniform
+// even when a group's row index runs past R.
+template <int G>
+ACCO_DEV float group_reduce_sum(float x, float* lds, int g, int lane) {
+  constexpr int WPG = (BLOCK / G) / 64;   // waves per group
   for (int off = 32; off > 0; off >>= 1)
     x += __shfl_down(x, off, 64);
-  const int wave = threadIdx.x / 64;
-  if ((threadIdx.x & 63) == 0) lds[wave] = x;
+  if ((lane & 63) == 0) lds[g * WPG + lane / 64] = x;
   __syncthreads();
-  float r = lds[0] + lds[1] + lds[2] + lds[3];
+  float r = 0.0f;
+#pragma unroll
+  for (int wv = 0; wv < WPG; ++wv) r += lds[g * WPG + wv];
   __syncthreads();
   return r;
 }
@@ -51,11 +60,12 @@ ACCO_DEV void store8(u16* p, const float* f) {
 }
 
 // ------------------------------------------------------------ RMSNorm fwd
-template <int CH>
+template <int CH, int G>
 __global__ __launch_bounds__(BLOCK)
 void rmsnorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
                         u16* __restrict__ y, float* __restrict__ rstd,
                         long long R, int D, float eps) {
+  constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];
   const int nv = D / VEC;
@@ -63,34 +73,39 @@ void rmsnorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
     reinterpret_cast<uint4*>(w_lds)[c] = reinterpret_cast<const uint4*>(w)[c];
   __syncthreads();
 
+  const int g = threadIdx.x / TPG;
+  const int lane = threadIdx.x % TPG;
   int cid[CH];
   bool act[CH];
   float wf[CH][VEC];
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
-    cid[j] = threadIdx.x + j * BLOCK;
+    cid[j] = lane + j * TPG;
     act[j] = cid[j] < nv;
     if (act[j]) load8(w_lds + cid[j] * VEC, wf[j]);
   }
 
-  for (long long row = blockIdx.x; row < R; row += gridDim.x) {
+  for (long long base = (long long)blockIdx.x * G; base < R;
+       base += (long long)gridDim.x * G) {
+    const long long row = base + g;
+    const bool live = row < R;
     const u16* xr = x + row * D;
     u16* yr = y + row * D;
     float xs[CH][VEC];
     float ssq = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         load8(xr + cid[j] * VEC, xs[j]);
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) ssq += xs[j][kk] * xs[j][kk];
       }
-    ssq = block_reduce_sum(ssq, lds);
+    ssq = group_reduce_sum<G>(ssq, lds, g, lane);
     const float r = rsqrtf(ssq / (float)D + eps);
-    if (threadIdx.x == 0 && rstd != nullptr) rstd[row] = r;
+    if (live && lane == 0 && rstd != nullptr) rstd[row] = r;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         float o[VEC];
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) o[kk] = xs[j][kk] * r * wf[j][kk];
@@ -101,13 +116,14 @@ void rmsnorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
 
 // ------------------------------------------------------------ RMSNorm bwd
 // dx = r*(dy*w) - x * r^3/D * sum(dy*w*x);  dw_col = sum_rows dy*x*r
-template <int CH>
+template <int CH, int G>
 __global__ __launch_bounds__(BLOCK)
 void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                         const u16* __restrict__ w,
                         const float* __restrict__ rstd, u16* __restrict__ dx,
-                        float* __restrict__ dw_part,  // [gridDim.x, D] fp32
+                        float* __restrict__ dw_part,  // [grid*G, D] fp32
                         long long R, int D) {
+  constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];
   const int nv = D / VEC;
@@ -115,39 +131,44 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
     reinterpret_cast<uint4*>(w_lds)[c] = reinterpret_cast<const uint4*>(w)[c];
   __syncthreads();
 
+  const int g = threadIdx.x / TPG;
+  const int lane = threadIdx.x % TPG;
   int cid[CH];
   bool act[CH];
   float wf[CH][VEC], dwacc[CH][VEC];
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
-    cid[j] = threadIdx.x + j * BLOCK;
+    cid[j] = lane + j * TPG;
     act[j] = cid[j] < nv;
     if (act[j]) load8(w_lds + cid[j] * VEC, wf[j]);
 #pragma unroll
     for (int kk = 0; kk < VEC; ++kk) dwacc[j][kk] = 0.0f;
   }
 
-  for (long long row = blockIdx.x; row < R; row += gridDim.x) {
+  for (long long base = (long long)blockIdx.x * G; base < R;
+       base += (long long)gridDim.x * G) {
+    const long long row = base + g;
+    const bool live = row < R;
     const u16* dyr = dy + row * D;
     const u16* xr = x + row * D;
     u16* dxr = dx + row * D;
-    const float r = rstd[row];
+    const float r = live ? rstd[row] : 0.0f;
     float xs[CH][VEC], ds[CH][VEC];
     float dot = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         load8(xr + cid[j] * VEC, xs[j]);
         load8(dyr + cid[j] * VEC, ds[j]);
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk)
           dot += ds[j][kk] * wf[j][kk] * xs[j][kk];
       }
-    dot = block_reduce_sum(dot, lds);
+    dot = group_reduce_sum<G>(dot, lds, g, lane);
     const float coef = r * r * r * dot / (float)D;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         float o[VEC];
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) {
@@ -157,10 +178,9 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
         store8(dxr + cid[j] * VEC, o);
       }
   }
-  // one coalesced partial row per block (atomicAdd on 2k fp32 addresses
-  // from 4k blocks serialized ~3x worse than the pre-fix kernel; the
-  // wrapper reduces the [grid, D] scratch with one tiny torch sum)
-  float* out_row = dw_part + (long long)blockIdx.x * D;
+  // one coalesced partial row per (block, group); groups that never saw a
+  // live row still write their zeros (the scratch is allocated w/ empty)
+  float* out_row = dw_part + ((long long)blockIdx.x * G + g) * D;
 #pragma unroll
   for (int j = 0; j < CH; ++j)
     if (act[j])
@@ -170,13 +190,14 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
 }
 
 // ---------------------------------------------------------- LayerNorm fwd
-template <int CH>
+template <int CH, int G>
 __global__ __launch_bounds__(BLOCK)
 void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
                           const u16* __restrict__ b, u16* __restrict__ y,
                           float* __restrict__ mean_out,
                           float* __restrict__ rstd_out,
                           long long R, int D, float eps) {
+  constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 wb_lds[];
   const int nv = D / VEC;
@@ -186,12 +207,14 @@ void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
   }
   __syncthreads();
 
+  const int g = threadIdx.x / TPG;
+  const int lane = threadIdx.x % TPG;
   int cid[CH];
   bool act[CH];
   float wf[CH][VEC], bf[CH][VEC];
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
-    cid[j] = threadIdx.x + j * BLOCK;
+    cid[j] = lane + j * TPG;
     act[j] = cid[j] < nv;
     if (act[j]) {
       load8(wb_lds + cid[j] * VEC, wf[j]);
@@ -199,37 +222,40 @@ void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
     }
   }
 
-  for (long long row = blockIdx.x; row < R; row += gridDim.x) {
+  for (long long base = (long long)blockIdx.x * G; base < R;
+       base += (long long)gridDim.x * G) {
+    const long long row = base + g;
+    const bool live = row < R;
     const u16* xr = x + row * D;
     u16* yr = y + row * D;
     float xs[CH][VEC];
     float sum = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         load8(xr + cid[j] * VEC, xs[j]);
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) sum += xs[j][kk];
       }
-    const float mean = block_reduce_sum(sum, lds) / (float)D;
+    const float mean = group_reduce_sum<G>(sum, lds, g, lane) / (float)D;
     float var = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j])
+      if (live && act[j])
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) {
           float d = xs[j][kk] - mean;
           var += d * d;
         }
-    var = block_reduce_sum(var, lds) / (float)D;
+    var = group_reduce_sum<G>(var, lds, g, lane) / (float)D;
     const float r = rsqrtf(var + eps);
-    if (threadIdx.x == 0) {
+    if (live && lane == 0) {
       if (mean_out) mean_out[row] = mean;
       if (rstd_out) rstd_out[row] = r;
     }
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         float o[VEC];
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk)
@@ -242,16 +268,17 @@ void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
 // ---------------------------------------------------------- LayerNorm bwd
 // xhat=(x-mean)*r; dyw=dy*w
 // dx = r*(dyw - mean(dyw) - xhat*mean(dyw*xhat)); dw=Σ dy*xhat; db=Σ dy
-template <int CH>
+template <int CH, int G>
 __global__ __launch_bounds__(BLOCK)
 void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                           const u16* __restrict__ w,
                           const float* __restrict__ mean_in,
                           const float* __restrict__ rstd_in,
                           u16* __restrict__ dx,
-                          float* __restrict__ dw_part,  // [grid, D]
-                          float* __restrict__ db_part,  // [grid, D]
+                          float* __restrict__ dw_part,  // [grid*G, D]
+                          float* __restrict__ db_part,  // [grid*G, D]
                           long long R, int D) {
+  constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];
   const int nv = D / VEC;
@@ -259,29 +286,34 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
     reinterpret_cast<uint4*>(w_lds)[c] = reinterpret_cast<const uint4*>(w)[c];
   __syncthreads();
 
+  const int g = threadIdx.x / TPG;
+  const int lane = threadIdx.x % TPG;
   int cid[CH];
   bool act[CH];
   float wf[CH][VEC], dwacc[CH][VEC], dbacc[CH][VEC];
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
-    cid[j] = threadIdx.x + j * BLOCK;
+    cid[j] = lane + j * TPG;
     act[j] = cid[j] < nv;
     if (act[j]) load8(w_lds + cid[j] * VEC, wf[j]);
 #pragma unroll
     for (int kk = 0; kk < VEC; ++kk) { dwacc[j][kk] = 0.f; dbacc[j][kk] = 0.f; }
   }
 
-  for (long long row = blockIdx.x; row < R; row += gridDim.x) {
+  for (long long base = (long long)blockIdx.x * G; base < R;
+       base += (long long)gridDim.x * G) {
+    const long long row = base + g;
+    const bool live = row < R;
     const u16* dyr = dy + row * D;
     const u16* xr = x + row * D;
     u16* dxr = dx + row * D;
-    const float mean = mean_in[row];
-    const float r = rstd_in[row];
+    const float mean = live ? mean_in[row] : 0.0f;
+    const float r = live ? rstd_in[row] : 0.0f;
     float xh[CH][VEC], ds[CH][VEC];
     float s1 = 0.0f, s2 = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         float xs[VEC];
         load8(xr + cid[j] * VEC, xs);
         load8(dyr + cid[j] * VEC, ds[j]);
@@ -293,12 +325,11 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
           s2 += dyw * xh[j][kk];
         }
       }
-    s1 = block_reduce_sum(s1, lds) / (float)D;
-    __syncthreads();
-    s2 = block_reduce_sum(s2, lds) / (float)D;
+    s1 = group_reduce_sum<G>(s1, lds, g, lane) / (float)D;
+    s2 = group_reduce_sum<G>(s2, lds, g, lane) / (float)D;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
-      if (act[j]) {
+      if (live && act[j]) {
         float o[VEC];
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) {
@@ -309,8 +340,8 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
         store8(dxr + cid[j] * VEC, o);
       }
   }
-  float* wrow = dw_part + (long long)blockIdx.x * D;
-  float* brow = db_part + (long long)blockIdx.x * D;
+  float* wrow = dw_part + ((long long)blockIdx.x * G + g) * D;
+  float* brow = db_part + ((long long)blockIdx.x * G + g) * D;
 #pragma unroll
   for (int j = 0; j < CH; ++j)
     if (act[j])
@@ -321,8 +352,12 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
       }
 }
 
-template <template <int> class K>
-struct ChDispatch {};
+int groups_for(int D) {
+  const int nv = D / VEC;
+  if (nv <= 64) return 4;      // one wave per row
+  if (nv <= 128) return 2;     // two waves per row (e.g. gptneo D=768)
+  return 1;
+}
 
 int chunks_for(int D) {
   const int nv = D / VEC;
@@ -332,22 +367,41 @@ int chunks_for(int D) {
   return 8;
 }
 
+int fwd_blocks(long long R, int G) {
+  long long b = (R + G - 1) / G;
+  const long long cap = 8192 / G;
+  return (int)((b < cap) ? (b < 1 ? 1 : b) : cap);
+}
+
+int bwd_blocks(long long R, int G) {
+  long long b = (R + G - 1) / G;
+  const long long cap = 2048 / G;
+  return (int)((b < cap) ? (b < 1 ? 1 : b) : cap);
+}
+
 }  // namespace
 
 extern "C" {
 
-int acco_norm_bwd_grid(long long R) { return (int)((R < 2048) ? R : 2048); }
+// number of fp32 partial rows the bwd kernels emit (scratch allocation)
+int acco_norm_bwd_grid(long long R, int D) {
+  const int G = groups_for(D);
+  return bwd_blocks(R, G) * G;
+}
 
 void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
                       long long R, int D, float eps, hipStream_t s) {
-  int grid = (int)((R < 8192) ? R : 8192);
   const int lds = D * sizeof(u16);
-#define L(CH) hipLaunchKernelGGL(rmsnorm_fwd_kernel<CH>, dim3(grid), \
+  const int G = groups_for(D);
+  const int grid = fwd_blocks(R, G);
+#define L(CH, G) hipLaunchKernelGGL((rmsnorm_fwd_kernel<CH, G>), dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)x, (const u16*)w, (u16*)y, \
     (float*)rstd, R, D, eps)
-  switch (chunks_for(D)) {
-    case 1: L(1); break; case 2: L(2); break;
-    case 4: L(4); break; default: L(8); break;
+  if (G == 4) { L(1, 4); }
+  else if (G == 2) { L(1, 2); }
+  else switch (chunks_for(D)) {
+    case 1: L(1, 1); break; case 2: L(2, 1); break;
+    case 4: L(4, 1); break; default: L(8, 1); break;
   }
 #undef L
 }
@@ -355,14 +409,17 @@ void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
 void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
                       const void* rstd, void* dx, void* dw_fp32,
                       long long R, int D, hipStream_t s) {
-  int grid = acco_norm_bwd_grid(R);
   const int lds = D * sizeof(u16);
-#define L(CH) hipLaunchKernelGGL(rmsnorm_bwd_kernel<CH>, dim3(grid), \
+  const int G = groups_for(D);
+  const int grid = bwd_blocks(R, G);
+#define L(CH, G) hipLaunchKernelGGL((rmsnorm_bwd_kernel<CH, G>), dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, (const u16*)w, \
     (const float*)rstd, (u16*)dx, (float*)dw_fp32, R, D)
-  switch (chunks_for(D)) {
-    case 1: L(1); break; case 2: L(2); break;
-    case 4: L(4); break; default: L(8); break;
+  if (G == 4) { L(1, 4); }
+  else if (G == 2) { L(1, 2); }
+  else switch (chunks_for(D)) {
+    case 1: L(1, 1); break; case 2: L(2, 1); break;
+    case 4: L(4, 1); break; default: L(8, 1); break;
   }
 #undef L
 }
@@ -370,14 +427,17 @@ void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
 void acco_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
                         void* mean, void* rstd, long long R, int D, float eps,
                         hipStream_t s) {
-  int grid = (int)((R < 8192) ? R : 8192);
   const int lds = 2 * D * sizeof(u16);
-#define L(CH) hipLaunchKernelGGL(layernorm_fwd_kernel<CH>, dim3(grid), \
-    dim3(BLOCK), lds, s, (const u16*)x, (const u16*)w, (const u16*)b, \
-    (u16*)y, (float*)mean, (float*)rstd, R, D, eps)
-  switch (chunks_for(D)) {
-    case 1: L(1); break; case 2: L(2); break;
-    case 4: L(4); break; default: L(8); break;
+  const int G = groups_for(D);
+  const int grid = fwd_blocks(R, G);
+#define L(CH, G) hipLaunchKernelGGL((layernorm_fwd_kernel<CH, G>), \
+    dim3(grid), dim3(BLOCK), lds, s, (const u16*)x, (const u16*)w, \
+    (const u16*)b, (u16*)y, (float*)mean, (float*)rstd, R, D, eps)
+  if (G == 4) { L(1, 4); }
+  else if (G == 2) { L(1, 2); }
+  else switch (chunks_for(D)) {
+    case 1: L(1, 1); break; case 2: L(2, 1); break;
+    case 4: L(4, 1); break; default: L(8, 1); break;
   }
 #undef L
 }
@@ -386,15 +446,18 @@ void acco_layernorm_bwd(const void* dy, const void* x, const void* w,
                         const void* mean, const void* rstd, void* dx,
                         void* dw_fp32, void* db_fp32, long long R, int D,
                         hipStream_t s) {
-  int grid = acco_norm_bwd_grid(R);
   const int lds = D * sizeof(u16);
-#define L(CH) hipLaunchKernelGGL(layernorm_bwd_kernel<CH>, dim3(grid), \
-    dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, (const u16*)w, \
-    (const float*)mean, (const float*)rstd, (u16*)dx, (float*)dw_fp32, \
-    (float*)db_fp32, R, D)
-  switch (chunks_for(D)) {
-    case 1: L(1); break; case 2: L(2); break;
-    case 4: L(4); break; default: L(8); break;
+  const int G = groups_for(D);
+  const int grid = bwd_blocks(R, G);
+#define L(CH, G) hipLaunchKernelGGL((layernorm_bwd_kernel<CH, G>), \
+    dim3(grid), dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, \
+    (const u16*)w, (const float*)mean, (const float*)rstd, (u16*)dx, \
+    (float*)dw_fp32, (float*)db_fp32, R, D)
+  if (G == 4) { L(1, 4); }
+  else if (G == 2) { L(1, 2); }
+  else switch (chunks_for(D)) {
+    case 1: L(1, 1); break; case 2: L(2, 1); break;
+    case 4: L(4, 1); break; default: L(8, 1); break;
   }
 #undef L
 }

@@ -47,7 +47,7 @@ def test_gelu_new_fwd_bwd():
     assert _close_bf16(x.grad, x32.grad)
 
 
-@pytest.mark.parametrize("D", [768, 2048])
+@pytest.mark.parametrize("D", [256, 768, 2048])
 def test_rmsnorm_fwd_bwd(D):
     from acco_amd import ops
     from acco_amd.ops import torch_ref
