@@ -376,7 +376,10 @@ int fwd_blocks(long long R, int G) {
 
 int bwd_blocks(long long R, int G) {
   long long b = (R + G - 1) / G;
-  const long long cap = 2048 / G;
+  // cap partial rows at 1024: the [partials, D] fp32 scratch write + torch
+  // reduce is ~25% of bwd traffic at 2048 partials and halves here, while
+  // 1024 blocks x 4 waves still fill all 1024 SIMDs
+  const long long cap = 1024 / G;
   return (int)((b < cap) ? (b < 1 ? 1 : b) : cap);
 }
 

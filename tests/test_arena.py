@@ -44,3 +44,24 @@ def test_flat_update_changes_forward():
     flat.mul_(0.5)
     y1 = m(x)
     assert not torch.allclose(y0, y1)
+
+
+def test_shard_spec_world8_llama1b_size():
+    """Driver runs N=1,2,4,8; verify the bucket-major layout invariants at
+    world=8 with the flagship parameter count."""
+    from acco_amd.parallel.comm import ShardSpec
+
+    n = 1_076_168_704          # llama-1b live numel (bench.py)
+    for world in (1, 2, 4, 8):
+        spec = ShardSpec.build(n, world, buckets=8)
+        assert spec.seg % 256 == 0
+        assert spec.total >= n
+        assert spec.total == spec.nb * world * spec.seg
+        # every (bucket, rank) segment contiguous and inside the buffer
+        last = 0
+        for b in range(spec.nb):
+            for r in range(world):
+                off = (b * world + r) * spec.seg
+                assert off == last
+                last = off + spec.seg
+        assert last == spec.total
