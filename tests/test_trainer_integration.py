@@ -186,3 +186,56 @@ def test_trainer_acco_finetune_ragged_ws2():
                       weights_only=False) for r in range(2)]
     assert torch.equal(res[0]["params"], res[1]["params"])
     assert torch.isfinite(res[0]["params"]).all()
+
+
+def _worker_eval(rank, world, port, tmpdir):
+    """Eval cadence (reference eval_loop :399-415 + train_acco eval at
+    :527-531): eval_loss computed every eval_step grads and logged."""
+    os.environ.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                       "RANK": str(rank), "LOCAL_RANK": str(rank),
+                       "WORLD_SIZE": str(world)})
+    os.chdir(tmpdir)
+    from acco_amd.config import load_config
+    from acco_amd.data.synthetic import SyntheticCausalLMDataset
+    from acco_amd.engine.trainer import DecoupledTrainer
+    from acco_amd.models import GPTNeoConfig, GPTNeoForCausalLM
+
+    cfg = load_config([
+        "train=acco", "train.nb_steps_tot=8", "train.batch_size=2",
+        "train.max_length=16", "train.use_mixed_precision=false",
+        "train.save=false", "train.eval=true", "train.eval_step=2",
+        "train.n_grad_accumulation=1", "train.comm_buckets=2",
+        "train.dataloader_num_workers=0",
+        "train.dataloader_persistent_workers=false",
+    ])
+    torch.manual_seed(42)
+    mcfg = GPTNeoConfig(hidden_size=32, num_layers=1, num_heads=2,
+                        vocab_size=64, max_position_embeddings=64,
+                        window_size=8)
+    model = GPTNeoForCausalLM(mcfg)
+
+    trainer = DecoupledTrainer(
+        model=model, tokenizer=None,
+        train_dataset=SyntheticCausalLMDataset(16, 16, 64, seed=3 + rank),
+        eval_dataset=SyntheticCausalLMDataset(4, 16, 64, seed=99),
+        args=cfg.train, run_name="evalrun")
+    evals = []
+    orig = trainer.eval_loop
+
+    def spy():
+        v = orig()
+        evals.append(v)
+        return v
+
+    trainer.eval_loop = spy
+    trainer.train()
+    torch.save({"n_evals": len(evals), "vals": evals},
+               os.path.join(tmpdir, f"ev_{rank}.pt"))
+    teardown_worker()
+
+
+def test_trainer_eval_cadence_ws2():
+    tmpdir = run_distributed(_worker_eval, 2, timeout=300)
+    res = torch.load(os.path.join(tmpdir, "ev_0.pt"), weights_only=False)
+    assert res["n_evals"] >= 1          # cadence fired (rank 0 logs it)
+    assert all(v == v and v > 0 for v in res["vals"])  # finite, positive
