@@ -19,6 +19,13 @@
 
 namespace {
 
+typedef float f4_ __attribute__((ext_vector_type(4)));
+typedef unsigned short u4_ __attribute__((ext_vector_type(4)));
+
+// p/m/v/g/out are each streamed exactly once per round: nontemporal
+// loads/stores keep the ~3.4 GB working set out of L2 (guide: streaming
+// kernels should not thrash caches shared with the concurrently running
+// compute stream).
 template <typename Tbuf, bool COMMIT>
 __global__ void fused_adamw_kernel(
     float* __restrict__ p, const Tbuf* __restrict__ g,
@@ -35,49 +42,76 @@ __global__ void fused_adamw_kernel(
   const long long stride = (long long)gridDim.x * blockDim.x;
   const long long n4 = n >> 2;
 
-  for (long long i = i0; i < n4; i += stride) {
-    float4 pf = reinterpret_cast<const float4*>(p)[i];
-    float4 mf = reinterpret_cast<const float4*>(m)[i];
-    float4 vf = reinterpret_cast<const float4*>(v)[i];
-    float gf[4];
-    if constexpr (sizeof(Tbuf) == 2) {
-      ushort4 gu = reinterpret_cast<const ushort4*>(g)[i];
-      gf[0] = bf16_to_f32(gu.x); gf[1] = bf16_to_f32(gu.y);
-      gf[2] = bf16_to_f32(gu.z); gf[3] = bf16_to_f32(gu.w);
-    } else {
-      float4 gv = reinterpret_cast<const float4*>(g)[i];
-      gf[0] = gv.x; gf[1] = gv.y; gf[2] = gv.z; gf[3] = gv.w;
-    }
+  // two grid-stride chunks per iteration: 8 independent loads in flight
+  // before any math retires (latency hiding for the HBM round trips)
+  auto body = [&](long long i, const f4_& pf, const f4_& mf, const f4_& vf,
+                  const float* gf) {
     float po[4], mo[4], vo[4];
-    float pp[4] = {pf.x, pf.y, pf.z, pf.w};
-    float mm[4] = {mf.x, mf.y, mf.z, mf.w};
-    float vv[4] = {vf.x, vf.y, vf.z, vf.w};
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
       float gk = gf[k] * s;
-      float pk = pp[k] * wd_factor;
-      float mk = beta1 * mm[k] + (1.0f - beta1) * gk;
-      float vk = beta2 * vv[k] + (1.0f - beta2) * gk * gk;
+      float pk = pf[k] * wd_factor;
+      float mk = beta1 * mf[k] + (1.0f - beta1) * gk;
+      float vk = beta2 * vf[k] + (1.0f - beta2) * gk * gk;
       float denom = sqrtf(vk) * inv_bc2_sqrt + eps;
       pk -= lr * inv_bc1 * mk / denom;
       po[k] = pk; mo[k] = mk; vo[k] = vk;
     }
     if constexpr (COMMIT) {
-      reinterpret_cast<float4*>(p)[i] = make_float4(po[0], po[1], po[2], po[3]);
-      reinterpret_cast<float4*>(m)[i] = make_float4(mo[0], mo[1], mo[2], mo[3]);
-      reinterpret_cast<float4*>(v)[i] = make_float4(vo[0], vo[1], vo[2], vo[3]);
+      __builtin_nontemporal_store(f4_{po[0], po[1], po[2], po[3]},
+                                  reinterpret_cast<f4_*>(p) + i);
+      __builtin_nontemporal_store(f4_{mo[0], mo[1], mo[2], mo[3]},
+                                  reinterpret_cast<f4_*>(m) + i);
+      __builtin_nontemporal_store(f4_{vo[0], vo[1], vo[2], vo[3]},
+                                  reinterpret_cast<f4_*>(v) + i);
     }
     if (out != nullptr) {
       if constexpr (sizeof(Tbuf) == 2) {
-        ushort4 ou;
-        ou.x = f32_to_bf16(po[0]); ou.y = f32_to_bf16(po[1]);
-        ou.z = f32_to_bf16(po[2]); ou.w = f32_to_bf16(po[3]);
-        reinterpret_cast<ushort4*>(out)[i] = ou;
+        u4_ ou = {f32_to_bf16(po[0]), f32_to_bf16(po[1]),
+                  f32_to_bf16(po[2]), f32_to_bf16(po[3])};
+        __builtin_nontemporal_store(ou, reinterpret_cast<u4_*>(out) + i);
       } else {
-        reinterpret_cast<float4*>(out)[i] =
-            make_float4(po[0], po[1], po[2], po[3]);
+        __builtin_nontemporal_store(f4_{po[0], po[1], po[2], po[3]},
+                                    reinterpret_cast<f4_*>(out) + i);
       }
     }
+  };
+  auto load_g = [&](long long i, float* gf) {
+    if constexpr (sizeof(Tbuf) == 2) {
+      u4_ gu = __builtin_nontemporal_load(
+          reinterpret_cast<const u4_*>(g) + i);
+      gf[0] = bf16_to_f32(gu[0]); gf[1] = bf16_to_f32(gu[1]);
+      gf[2] = bf16_to_f32(gu[2]); gf[3] = bf16_to_f32(gu[3]);
+    } else {
+      f4_ gv = __builtin_nontemporal_load(
+          reinterpret_cast<const f4_*>(g) + i);
+      gf[0] = gv[0]; gf[1] = gv[1]; gf[2] = gv[2]; gf[3] = gv[3];
+    }
+  };
+
+  long long i = i0;
+  for (; i + stride < n4; i += 2 * stride) {
+    const long long i2 = i + stride;
+    f4_ pf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(p) + i);
+    f4_ mf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(m) + i);
+    f4_ vf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(v) + i);
+    float gf[4];
+    load_g(i, gf);
+    f4_ pf2 = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(p) + i2);
+    f4_ mf2 = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(m) + i2);
+    f4_ vf2 = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(v) + i2);
+    float gf2[4];
+    load_g(i2, gf2);
+    body(i, pf, mf, vf, gf);
+    body(i2, pf2, mf2, vf2, gf2);
+  }
+  for (; i < n4; i += stride) {
+    f4_ pf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(p) + i);
+    f4_ mf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(m) + i);
+    f4_ vf = __builtin_nontemporal_load(reinterpret_cast<const f4_*>(v) + i);
+    float gf[4];
+    load_g(i, gf);
+    body(i, pf, mf, vf, gf);
   }
 
   // scalar tail (segments are 256-aligned so this is normally empty)
