@@ -63,10 +63,15 @@ def _worker_acco(rank, world, port, tmpdir, n_warmup):
     sched = LRSchedule(1e-2, 4, TARGET, "cosine")
 
     batches = make_batches(rank)
-    it = iter(batches)
+    bptr = [0]
 
     def next_batch():
-        return next(it)
+        # cycle: under heavy host load the compute thread can accumulate
+        # arbitrarily many micro-batches per com round (ACCO heterogeneity);
+        # the oracle's take() cycles with the same modulo
+        b = batches[bptr[0] % len(batches)]
+        bptr[0] += 1
+        return b
 
     def forward_backward(batch):
         x, y = batch
@@ -113,7 +118,7 @@ def oracle_replay(w0, traces, world, n_warmup, lr=1e-2):
     ptr = {r: 0 for r in range(world)}
 
     def take(r):
-        b = batches[r][ptr[r]]
+        b = batches[r][ptr[r] % len(batches[r])]
         ptr[r] += 1
         return b
 

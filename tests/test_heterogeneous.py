@@ -30,7 +30,14 @@ def _worker_hetero(rank, world, port, tmpdir):
     comm = CommBackend(device)
     opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
     sched = LRSchedule(1e-2, 4, TARGET, "cosine")
-    it = iter(make_batches(rank))
+    batches = make_batches(rank)
+    bptr = [0]
+
+    def next_batch():
+        # cycle — compute can outpace com arbitrarily under host load
+        b = batches[bptr[0] % len(batches)]
+        bptr[0] += 1
+        return b
 
     def forward_backward(batch):
         x, y = batch
@@ -42,7 +49,7 @@ def _worker_hetero(rank, world, port, tmpdir):
     eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
                      spec=spec, comm=comm, rank=rank, device=device, opt=opt,
                      sched=sched, forward_backward=forward_backward,
-                     next_batch=lambda: next(it), n_grad_accumulation=n_acc)
+                     next_batch=next_batch, n_grad_accumulation=n_acc)
     opt.init_master_from_buffer(params)
     eng.trace = []
     eng.train_acco(TARGET, n_warmup_steps=0)
