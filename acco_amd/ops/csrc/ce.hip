@@ -41,12 +41,17 @@ __global__ void ce_fwd_kernel(const u16* __restrict__ logits,  // [T, V]
       float f[8] = {bf16_to_f32(a.x), bf16_to_f32(a.y), bf16_to_f32(a.z),
                     bf16_to_f32(a.w), bf16_to_f32(b.x), bf16_to_f32(b.y),
                     bf16_to_f32(b.z), bf16_to_f32(b.w)};
+      // chunk max first, one rescale of the running sum per 8 elements —
+      // replaces the per-element branchy online update (8 dependent
+      // branch+rescale chains) with independent exps
+      float cm = f[0];
 #pragma unroll
-      for (int k = 0; k < 8; ++k) {
-        float x = f[k];
-        if (x > m) { sum *= __expf(m - x); m = x; }
-        sum += __expf(x - m);
-      }
+      for (int k = 1; k < 8; ++k) cm = fmaxf(cm, f[k]);
+      const float mn = fmaxf(m, cm);
+      sum *= __expf(m - mn);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) sum += __expf(f[k] - mn);
+      m = mn;
     }
     for (int tail = nv * 8 + threadIdx.x; tail < V; tail += BLOCK) {
       float x = bf16_to_f32(row[tail]);
