@@ -12,6 +12,10 @@ would have put the per-projection grads).
 
 This is an MI355X-first arena dividend: bigger GEMMs fill the 256-CU chip
 (guide: a launch needs >>256 workgroups) and halve kernel-launch count.
+It subsumes the reference's per-projection GEMMs + AccumulateGrad adds on
+the flat grad vector (reference trainer_base.py:284-320 re-points
+param.grad into the flat tensor; autograd then adds each wgrad into it —
+here the wgrad lands in the arena in one step).
 """
 
 from __future__ import annotations
