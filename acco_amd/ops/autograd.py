@@ -127,8 +127,14 @@ class CausalLMLossFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         logits, labels, lse, acc = ctx.saved_tensors
-        dlogits = ops.hip_ext().ce_bwd(logits, labels, lse, acc,
-                                       float(grad_out))
+        ext = ops.hip_ext()
+        if grad_out.is_cuda and hasattr(ext, "ce_bwd_dev"):
+            # read the upstream grad on-device — float(grad_out) would be a
+            # D2H sync stalling the backward launch pipeline every micro
+            dlogits = ext.ce_bwd_dev(logits, labels, lse, acc,
+                                     grad_out.reshape(1).float())
+        else:
+            dlogits = ext.ce_bwd(logits, labels, lse, acc, float(grad_out))
         return dlogits, None
 
 

@@ -36,8 +36,8 @@ extern "C" void acco_rope(const void*, void*, const float*, const float*,
 extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
                             long long, int, int, hipStream_t);
 extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
-                            void*, const float*, float, long long, int, int,
-                            hipStream_t);
+                            void*, const float*, float, const float*,
+                            long long, int, int, hipStream_t);
 extern "C" void acco_gemm_nt(const void*, const void*, void*, int, int,
                              int, hipStream_t);
 extern "C" void acco_attn_fwd(const void*, const void*, const void*, void*,
@@ -280,7 +280,26 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
   acco_ce_bwd(logits.data_ptr(),
               reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
               lse.data_ptr<float>(), dlogits.data_ptr(),
-              acc.data_ptr<float>(), (float)dloss, T, S, V, cur_stream());
+              acc.data_ptr<float>(), (float)dloss, nullptr, T, S, V,
+              cur_stream());
+  return dlogits;
+}
+
+// dloss read from a 1-elem fp32 device tensor: no D2H sync in backward
+at::Tensor ce_bwd_dev(at::Tensor logits, at::Tensor labels, at::Tensor lse,
+                      at::Tensor acc, at::Tensor dloss_dev) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(dloss_dev.is_cuda() && dloss_dev.scalar_type() == at::kFloat &&
+              dloss_dev.numel() == 1);
+  const long long Bn = logits.size(0);
+  const int S = (int)logits.size(1), V = (int)logits.size(2);
+  const long long T = Bn * S;
+  auto dlogits = at::empty_like(logits);
+  acco_ce_bwd(logits.data_ptr(),
+              reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
+              lse.data_ptr<float>(), dlogits.data_ptr(),
+              acc.data_ptr<float>(), 0.0f, dloss_dev.data_ptr<float>(), T, S,
+              V, cur_stream());
   return dlogits;
 }
 
@@ -429,6 +448,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("ce_bwd_dev", &ce_bwd_dev);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);

@@ -95,8 +95,10 @@ __global__ void ce_bwd_kernel(const u16* __restrict__ logits,
                               u16* __restrict__ dlogits,
                               const float* __restrict__ acc,  // n_valid at [1]
                               float dloss,                    // upstream grad
+                              const float* __restrict__ dloss_dev,  // device override (avoids a D2H sync in backward)
                               long long T, int S, int V) {
-  const float scale = dloss / fmaxf(acc[1], 1.0f);
+  const float dl = (dloss_dev != nullptr) ? *dloss_dev : dloss;
+  const float scale = dl / fmaxf(acc[1], 1.0f);
   for (long long t = blockIdx.x; t < T; t += gridDim.x) {
     const int s_pos = (int)(t % S);
     long long label = -100;
@@ -153,11 +155,12 @@ void acco_ce_fwd(const void* logits, const long long* labels, float* lse,
 
 void acco_ce_bwd(const void* logits, const long long* labels,
                  const float* lse, void* dlogits, const float* acc,
-                 float dloss, long long T, int S, int V, hipStream_t s) {
+                 float dloss, const float* dloss_dev, long long T, int S,
+                 int V, hipStream_t s) {
   int grid = (int)((T < 8192) ? T : 8192);
   hipLaunchKernelGGL(ce_bwd_kernel, dim3(grid), dim3(BLOCK), 0, s,
                      (const u16*)logits, labels, lse, (u16*)dlogits, acc,
-                     dloss, T, S, V);
+                     dloss, dloss_dev, T, S, V);
 }
 
 }  // extern "C"
