@@ -343,14 +343,13 @@ class DecoupledTrainer:
         count_grad_tot = 0
         count_com = 0
         n_acc = int(self.args.n_grad_accumulation)
-        loss_val = 0.0
+        loss_last = None
         while count_grad_tot < self.nb_grad_tot:
             for step in range(n_acc):
                 if step == n_acc - 1:
                     self.ddp.begin_sync_microbatch()
                 inputs = self.load_next_batch()
-                loss = self.forward_backward(inputs)
-                loss_val = float(loss.float().item())
+                loss_last = self.forward_backward(inputs)
             lr = self.sched.lr()
             self.ddp.finish_step(grad_scale=1.0 / self.world_size, lr=lr)
             self.ddp.zero_grad()
@@ -358,6 +357,9 @@ class DecoupledTrainer:
             count_grad_tot += self.world_size * n_acc
             count_com += 1
             if self.rank == 0:
+                # read the displayed loss once per com round (a per-micro
+                # .item() would D2H-sync inside the accumulation loop)
+                loss_val = float(loss_last.float().item())
                 eval_loss = None
                 if self.args.eval and self.eval_dataloader is not None:
                     if count_grad_tot - self._last_eval > self.args.eval_step:
