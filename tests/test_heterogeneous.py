@@ -76,3 +76,20 @@ def test_acco_heterogeneous_accumulation_ws2():
     P, count_tot = oracle_replay(w0, traces, 2, 0)
     assert count_tot == res[0]["count"]
     assert torch.allclose(P, res[0]["params"], atol=1e-6, rtol=1e-6)
+
+
+def _worker_handoff(rank, world, port, tmpdir):
+    """ACCO_DEBUG_HANDOFF=1: the com-buffer ownership checksum must hold
+    across every comm->compute handoff (SURVEY.md §5 race-detection mode)."""
+    os.environ["ACCO_DEBUG_HANDOFF"] = "1"
+    try:
+        _worker_hetero(rank, world, port, tmpdir)
+    finally:
+        os.environ.pop("ACCO_DEBUG_HANDOFF", None)
+
+
+def test_acco_handoff_checksum_ws2():
+    tmpdir = run_distributed(_worker_handoff, 2, timeout=240)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(2)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
