@@ -27,7 +27,7 @@ def test_save_result_schema_union(tmp_path):
 def test_create_dict_result_fields():
     row = create_dict_result({"learning_rate": 1e-4, "method_name": "acco"},
                              world_size=8, n_nodes=1, device_name="MI355X",
-                             total_time=12.5, id_run="xyz", loss=2.25)
+                             total_time=12.5, id_run="xyz", loss_final=2.25)
     assert row["N_workers"] == 8
     assert row["cuda_device"] == "MI355X"
     assert row["Tot_time"] == 12.5
