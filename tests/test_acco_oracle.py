@@ -217,7 +217,12 @@ def _run_mode(rank, world, tmpdir, mode):
     opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
     sched = LRSchedule(1e-2, 4, TARGET, "cosine")
     batches = make_batches(rank)
-    it = iter(batches)
+    bptr = [0]
+
+    def next_batch():
+        b = batches[bptr[0] % len(batches)]     # cycle (see _worker_acco)
+        bptr[0] += 1
+        return b
 
     def forward_backward(batch):
         x, y = batch
@@ -228,7 +233,7 @@ def _run_mode(rank, world, tmpdir, mode):
     eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
                      spec=spec, comm=comm, rank=rank, device=device, opt=opt,
                      sched=sched, forward_backward=forward_backward,
-                     next_batch=lambda: next(it), n_grad_accumulation=N_ACC)
+                     next_batch=next_batch, n_grad_accumulation=N_ACC)
     opt.init_master_from_buffer(params)
     if mode == "dpu":
         eng.train_dpu(TARGET, n_warmup_steps=0)
@@ -268,7 +273,13 @@ def _run_warmup(rank, world, tmpdir):
     comm = CommBackend(device)
     opt = ShardedAdamW(spec, rank, device, lr=1e-2, weight_decay=0.01)
     sched = LRSchedule(1e-2, 4, TARGET, "cosine")
-    it = iter(make_batches(rank))
+    batches = make_batches(rank)
+    bptr = [0]
+
+    def next_batch():
+        b = batches[bptr[0] % len(batches)]     # cycle (see _worker_acco)
+        bptr[0] += 1
+        return b
 
     def forward_backward(batch):
         x, y = batch
@@ -279,7 +290,7 @@ def _run_warmup(rank, world, tmpdir):
     eng = AccoEngine(params_arena=params, grads_arena=grads, n_live=D,
                      spec=spec, comm=comm, rank=rank, device=device, opt=opt,
                      sched=sched, forward_backward=forward_backward,
-                     next_batch=lambda: next(it), n_grad_accumulation=N_ACC)
+                     next_batch=next_batch, n_grad_accumulation=N_ACC)
     opt.init_master_from_buffer(params)
     eng.train_acco(TARGET, n_warmup_steps=3)
     torch.save({"params": params[:D].clone(), "count": eng.count_grad_tot},
