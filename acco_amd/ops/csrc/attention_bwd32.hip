@@ -1,6 +1,8 @@
 // Flash attention backward v4 for gfx950 — 32×32 MFMA structure with
 // in-register P/dS redistribution (no per-wave LDS round trips), mirroring
-// attention_fwd32.hip. D = 64 and S % 256 == 0 only (v3 fallback else).
+// attention_fwd32.hip. D ∈ {64, 128} and S % 256 == 0 (v3 fallback else);
+// the D=128 dkv build trades K/V register residency and T14 staging for
+// accumulator headroom (see KV_RES) — 256 VGPRs, 6 spilled, occupancy 2.
 //
 //   dq kernel : grid over 256-row Q blocks (8 waves × 32 q rows);
 //               per 64-kv tile recompute S^T = K·Q^T and dP^T = V·dO^T
@@ -43,6 +45,24 @@ ACCO_DEV short8 c_to_afrag(const float* c, int kk) {
   unsigned su1 = pack_bf16_(pp[2], pp[3]);
   unsigned su2 = pack_bf16_(pp[4], pp[5]);
   unsigned su3 = pack_bf16_(pp[6], pp[7]);
+  auto r02 = __builtin_amdgcn_permlane32_swap(su0, su2, false, false);
+  auto r13 = __builtin_amdgcn_permlane32_swap(su1, su3, false, false);
+  short8 pa;
+  reinterpret_cast<unsigned*>(&pa)[0] = r02[0];
+  reinterpret_cast<unsigned*>(&pa)[1] = r13[0];
+  reinterpret_cast<unsigned*>(&pa)[2] = r02[1];
+  reinterpret_cast<unsigned*>(&pa)[3] = r13[1];
+  return pa;
+}
+
+// same redistribution over an f32x16 C fragment kept in vector registers
+// (constant kk from an unrolled loop — element reads stay in VGPRs)
+ACCO_DEV short8 c_to_afrag_v(const f32x16& c, int kk) {
+  const int b = kk * 8;
+  unsigned su0 = pack_bf16_(c[b + 0], c[b + 1]);
+  unsigned su1 = pack_bf16_(c[b + 2], c[b + 3]);
+  unsigned su2 = pack_bf16_(c[b + 4], c[b + 5]);
+  unsigned su3 = pack_bf16_(c[b + 6], c[b + 7]);
   auto r02 = __builtin_amdgcn_permlane32_swap(su0, su2, false, false);
   auto r13 = __builtin_amdgcn_permlane32_swap(su1, su3, false, false);
   short8 pa;
@@ -234,14 +254,20 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   const u16* Qb = q + (long long)b * S * qs + (long long)h * D;
   const u16* dOb = dO + (long long)b * S * do_rs + (long long)h * D;
 
-  // K^T / V^T as B operands: lane = K[kv=lq][d=hi*8+i+16s]
-  short8 kTf[KS], vTf[KS];
+  // K^T / V^T as B operands: lane = K[kv=lq][d=hi*8+i+16s].
+  // Register-resident at D=64 (64 VGPRs); at D=128 that residency pushes
+  // the kernel to 91 spilled VGPRs, so the fragments are re-read from the
+  // L2-hot global rows inside the K-step loop instead.
+  constexpr bool KV_RES = (D == 64);
+  short8 kTf[KV_RES ? KS : 1], vTf[KV_RES ? KS : 1];
+  if constexpr (KV_RES) {
 #pragma unroll
-  for (int s = 0; s < KS; ++s) {
-    kTf[s] = *reinterpret_cast<const short8*>(
-        Kp + (long long)lq * ks + s * 16 + hi * 8);
-    vTf[s] = *reinterpret_cast<const short8*>(
-        Vp + (long long)lq * ks + s * 16 + hi * 8);
+    for (int s = 0; s < KS; ++s) {
+      kTf[s] = *reinterpret_cast<const short8*>(
+          Kp + (long long)lq * ks + s * 16 + hi * 8);
+      vTf[s] = *reinterpret_cast<const short8*>(
+          Vp + (long long)lq * ks + s * 16 + hi * 8);
+    }
   }
 
   f32x16 acc_dk[DT], acc_dv[DT];
@@ -314,12 +340,36 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
   };
 
-  load_qtile(qt_lo);
+  // direct (non-T14) staging for the !KV_RES (D=128) build: the
+  // cross-barrier register staging holds 32 VGPRs for the whole loop and
+  // tips the 128-VGPR accumulator build into spilling
+  auto stage_qtile_direct = [&](int qt) {
+    const u16* Qs = Qb + (long long)(qt * KT) * qs;
+    const u16* Ds = dOb + (long long)(qt * KT) * do_rs;
+    stage_T<D>(Qs, qs, qT_lds);
+    stage_T<D>(Ds, do_rs, doT_lds);
+#pragma unroll
+    for (int c = 0; c < RCH; ++c) {
+      const int cc = threadIdx.x + c * 512;
+      const int r = cc / (D / 8), dc = cc % (D / 8);
+      reinterpret_cast<uint4*>(q_row + r * KROW)[dc] =
+          *reinterpret_cast<const uint4*>(Qs + (long long)r * qs + dc * 8);
+      reinterpret_cast<uint4*>(do_row + r * KROW)[dc] =
+          *reinterpret_cast<const uint4*>(Ds + (long long)r * do_rs + dc * 8);
+    }
+  };
+
+  if constexpr (KV_RES) load_qtile(qt_lo);
   for (int qt = qt_lo; qt <= qt_hi; ++qt) {
     __syncthreads();
-    write_qtile();
+    if constexpr (KV_RES) {
+      write_qtile();
+    } else {
+      stage_qtile_direct(qt);
+    }
     __syncthreads();
-    if (qt < qt_hi) load_qtile(qt + 1);  // in flight under the MFMAs
+    if constexpr (KV_RES)
+      if (qt < qt_hi) load_qtile(qt + 1);  // in flight under the MFMAs
     // tile fully before this wave's kv rows → all masked: skip compute
     if (qt * KT + KT - 1 < kv_wave_min) continue;
 
@@ -333,20 +383,40 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 #pragma unroll
       for (int r = 0; r < 16; ++r) { st[r] = 0.0f; dpt[r] = 0.0f; }
       __builtin_amdgcn_s_setprio(1);
+      if constexpr (KV_RES) {
 #pragma unroll
-      for (int s = 0; s < KS; ++s) {
-        short8 qfr = *reinterpret_cast<const short8*>(
-            q_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
-        short8 dofr = *reinterpret_cast<const short8*>(
-            do_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
-        st = MFMA32(qfr, kTf[s], st);
-        dpt = MFMA32(dofr, vTf[s], dpt);
+        for (int s = 0; s < KS; ++s) {
+          short8 qfr = *reinterpret_cast<const short8*>(
+              q_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
+          short8 dofr = *reinterpret_cast<const short8*>(
+              do_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
+          st = MFMA32(qfr, kTf[s], st);
+          dpt = MFMA32(dofr, vTf[s], dpt);
+        }
+      } else {
+        // non-resident K/V (D=128): bounded unroll keeps the in-flight
+        // global loads from ballooning the register file (full unroll of
+        // 8 K-steps spills)
+#pragma unroll 2
+        for (int s = 0; s < KS; ++s) {
+          short8 qfr = *reinterpret_cast<const short8*>(
+              q_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
+          short8 dofr = *reinterpret_cast<const short8*>(
+              do_row + (m32 * 32 + lq) * KROW + s * 16 + hi * 8);
+          short8 kf = *reinterpret_cast<const short8*>(
+              Kp + (long long)lq * ks + s * 16 + hi * 8);
+          short8 vf = *reinterpret_cast<const short8*>(
+              Vp + (long long)lq * ks + s * 16 + hi * 8);
+          st = MFMA32(qfr, kf, st);
+          dpt = MFMA32(dofr, vf, dpt);
+        }
       }
       __builtin_amdgcn_s_setprio(0);
 
-      // P and dS (C: col = kv = lq, row = q spread)
+      // P and dS (C: col = kv = lq, row = q spread) — written back into
+      // st/dpt in place: separate p16/ds16 arrays cost 32 VGPRs and spill
+      // the D=128 instantiation
       const int kv_g = kv0 + lq;
-      float p16[16], ds16[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int rr = (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -356,16 +426,19 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
         bool valid = (kv_g <= q_g) && (q_g < S);
         if (window > 0) valid = valid && (kv_g > q_g - window);
         const float pval = valid ? __expf(st[r] * scale - lse_q) : 0.0f;
-        p16[r] = pval;
-        ds16[r] = pval * (dpt[r] - del_q) * scale;
+        st[r] = pval;
+        dpt[r] = pval * (dpt[r] - del_q) * scale;
       }
 
       // dV += P^T·dO ; dK += dS^T·Q   (A rows = kv via swap; B from LDS^T)
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
-        short8 pa = c_to_afrag(p16, kk);
-        short8 dsa = c_to_afrag(ds16, kk);
+        short8 pa = c_to_afrag_v(st, kk);
+        short8 dsa = c_to_afrag_v(dpt, kk);
+        // NOTE: must stay fully unrolled — t indexes the accumulator
+        // register arrays (guide rule 20: runtime indexing demotes them
+        // to scratch; a partial-unroll experiment cost 576 B/lane)
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
           short8 dob = *reinterpret_cast<const short8*>(
@@ -425,10 +498,16 @@ void acco_attn_bwd32_dkv(const void* q, const void* k, const void* v,
                          long long do_rs, hipStream_t stream) {
   dim3 grid(S / BT, B * H);
   const int lds = (2 * KT * (D + 8) + 2 * D * LST) * sizeof(u16);
-  hipLaunchKernelGGL(attn_bwd32_dkv_kernel<64>, grid, dim3(512), lds, stream,
-                     (const u16*)q, (const u16*)k, (const u16*)v,
-                     (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
-                     Hkv, scale, window, q_rs, kv_rs, do_rs);
+  if (D == 64)
+    hipLaunchKernelGGL(attn_bwd32_dkv_kernel<64>, grid, dim3(512), lds,
+                       stream, (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
+                       Hkv, scale, window, q_rs, kv_rs, do_rs);
+  else
+    hipLaunchKernelGGL(attn_bwd32_dkv_kernel<128>, grid, dim3(512), lds,
+                       stream, (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dO, lse, delta, (u16*)dk, (u16*)dv, S, H,
+                       Hkv, scale, window, q_rs, kv_rs, do_rs);
 }
 
 }  // extern "C"

@@ -395,7 +395,7 @@ void acco_attn_bwd_dkv(const void* q, const void* k, const void* v,
                        const void* dO, const float* lse, const float* delta,
                        void* dk, void* dv, int B, int S, int H, int Hkv,
                        int D, float scale, int window, hipStream_t stream) {
-  if (D == 64 && S % 256 == 0) {
+  if ((D == 64 || D == 128) && S % 256 == 0) {
     acco_attn_bwd32_dkv(q, k, v, dO, lse, delta, dk, dv, B, S, H, Hkv, D,
                         scale, window, (long long)H * D, (long long)Hkv * D,
                         (long long)H * D, stream);
