@@ -53,7 +53,7 @@ class LlamaAttention(nn.Module):
             from acco_amd.models.fuse import FusedArenaLinearFn
             w, g, splits = self._fused_qkv
             qkv = FusedArenaLinearFn.apply(x, w, g)
-            if (qkv.is_cuda and cfg.head_dim == 64 and S % 256 == 0
+            if (qkv.is_cuda and cfg.head_dim in (64, 128) and S % 256 == 0
                     and ops.have_kernel("attn_fwd_packed")):
                 # packed core: RoPE + flash attention directly on the fused
                 # projection output, packed grad back — zero split/cat

@@ -392,7 +392,7 @@ std::vector<at::Tensor> attn_fwd_packed(at::Tensor qkv, int64_t H,
   CHECK_BF16_CONTIG(qkv);
   const int B = (int)qkv.size(0), S = (int)qkv.size(1);
   const long long W = qkv.size(2);
-  TORCH_CHECK(W >= (H + 2 * Hkv) * D && D == 64 && S % 256 == 0);
+  TORCH_CHECK(W >= (H + 2 * Hkv) * D && (D == 64 || D == 128) && S % 256 == 0);
   const u16* base = (const u16*)qkv.data_ptr();
   auto o = at::empty({B, S, H * D}, qkv.options());
   auto lse = at::empty({B, H, S}, qkv.options().dtype(at::kFloat));

@@ -149,3 +149,41 @@ def test_packed_kvq_core_gptneo_matches_standard():
     cos = torch.nn.functional.cosine_similarity(g1[:n].float(),
                                                 g2[:n].float(), dim=0)
     assert cos > 0.995, f"grad cosine {cos}"
+
+
+@pytest.mark.gpu
+def test_packed_qkv_core_llama_d128_matches_standard():
+    """Packed attention core at head_dim=128 (llama-8b geometry): fused
+    projection + packed rope/attention/grad vs the standard path."""
+    import torch
+    from acco_amd.engine import arena
+    from acco_amd.models import LlamaConfig, LlamaForCausalLM
+    from acco_amd.models.fuse import install_fused_projections
+
+    cfg = LlamaConfig(hidden_size=256, num_layers=2, num_heads=2,
+                      num_kv_heads=1, intermediate_size=512, vocab_size=512,
+                      max_position_embeddings=512)
+    assert cfg.head_dim == 128
+    torch.manual_seed(0)
+    m1 = LlamaForCausalLM(cfg)
+    torch.manual_seed(0)
+    m2 = LlamaForCausalLM(cfg)
+
+    dev = torch.device("cuda")
+    p1 = arena.flatten_params(m1, torch.bfloat16, dev, pad_to=256)
+    g1 = arena.attach_grad_arena(m1, torch.bfloat16, dev, pad_to=256)
+    p2 = arena.flatten_params(m2, torch.bfloat16, dev, pad_to=256)
+    g2 = arena.attach_grad_arena(m2, torch.bfloat16, dev, pad_to=256)
+    assert torch.equal(p1, p2)
+    assert install_fused_projections(m2, p2, g2) > 0
+
+    ids = torch.randint(0, 512, (2, 256), device=dev)
+    loss1, _ = m1(ids, labels=ids)
+    loss1.backward()
+    loss2, _ = m2(ids, labels=ids)
+    loss2.backward()
+    assert abs(float(loss1) - float(loss2)) < 2e-2, (loss1, loss2)
+    n = arena.live_numel(m1)
+    cos = torch.nn.functional.cosine_similarity(g1[:n].float(),
+                                                g2[:n].float(), dim=0)
+    assert cos > 0.995, f"grad cosine {cos}"
