@@ -99,3 +99,20 @@ def test_load_pretrained_file_and_dir(tmp_path):
     m2 = LlamaForCausalLM(cfg)
     load_pretrained(m2, str(d))
     assert torch.equal(m2.lm_head.weight, src.lm_head.weight)
+
+
+def test_resume_end_to_end_cpu():
+    """Full train -> save -> fresh trainer -> restore -> continue cycle
+    (the scripted GPU check benchmarks/resume_check.py, CPU mode)."""
+    import subprocess
+    import sys
+    import tempfile
+
+    out = subprocess.run(
+        [sys.executable,
+         os.path.join(os.path.dirname(os.path.dirname(
+             os.path.abspath(__file__))), "benchmarks", "resume_check.py")],
+        capture_output=True, text=True, timeout=300,
+        cwd=tempfile.mkdtemp(prefix="acco_resume_test_"))
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "RESUME_OK" in out.stdout, out.stdout[-500:]
