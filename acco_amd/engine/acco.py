@@ -297,9 +297,14 @@ class AccoEngine:
         count_after_init=-1, i.e. commit+schedule every round, while
         update_buffers_step alternates grad zeroing)."""
         self.bootstrap(n_warmup_steps)
-        count_com = 0
+        # the grad-zeroing alternation continues from the warmup rounds'
+        # parity (reference initializes its round counter to n_warmup_steps,
+        # trainer_decoupled.py:618,659 — with an odd warmup the first
+        # threaded round must NOT zero)
+        count_com = n_warmup_steps
+        max_com = None if max_rounds is None else n_warmup_steps + max_rounds
         while self.count_grad_tot < nb_grad_tot and (
-                max_rounds is None or count_com < max_rounds):
+                max_com is None or count_com < max_com):
             for _ in range(self.n_acc):
                 self.gradient_step(self.next_batch())
             self.communication_round(commit=True)

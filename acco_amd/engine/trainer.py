@@ -76,14 +76,13 @@ class DecoupledTrainer:
         # C1: average random-init weights across ranks (reference :180)
         self.comm.all_reduce_avg(self.params)
 
-        # fused QKV / gate-up GEMMs over arena-adjacent weight views.
-        # Not in DDP mode: fused grads bypass per-param AccumulateGrad, and
-        # the native DDP's backward-overlap hooks rely on it (the reference
-        # DDP baseline also runs unfused projections).
-        if args.method_name != "ddp" and not getattr(args, "run_baseline_ddp",
-                                                     False):
-            from acco_amd.models.fuse import install_fused_projections
-            install_fused_projections(model, self.params, self.grads)
+        # fused QKV / gate-up GEMMs over arena-adjacent weight views — in
+        # EVERY mode, including DDP: the native DDP's bucket readiness is
+        # element-coverage based (parallel/ddp.py) so the in-place-dW path
+        # feeds its backward-overlap launches too. ACCO vs DDP therefore
+        # compares the algorithms on identical compute kernels.
+        from acco_amd.models.fuse import install_fused_projections
+        install_fused_projections(model, self.params, self.grads)
 
         # ---- observability
         out_dir = os.getcwd()

@@ -32,12 +32,32 @@ import torch.nn as nn
 # beta=0, outweighing the saved add kernel — profiles/ r01 v6 vs v7).
 _EPILOGUE = os.environ.get("ACCO_WGRAD_EPILOGUE", "0") == "1"
 
+# Grad-producer notifier: when armed (NativeZeroDDP arms it for the last
+# micro-batch of an accumulation window), the arena autograd Functions call
+# it with each grad-arena view right after accumulating into it, so the DDP
+# readiness tracker can launch a bucket's reduce-scatter as soon as every
+# producer intersecting the bucket has landed — the in-place-dW equivalent
+# of a post-accumulate-grad hook (those never fire for arena views, since
+# the weight *view* passed to the Function is not the leaf Parameter).
+_grad_notifier = None
+
+
+def set_grad_notifier(fn) -> None:
+    global _grad_notifier
+    _grad_notifier = fn
+
+
+def _notify(g_view) -> None:
+    if _grad_notifier is not None:
+        _grad_notifier(g_view.storage_offset(), g_view.numel())
+
 
 def _acc_wgrad(g_view, d2t, x2):
     if _EPILOGUE:
         g_view.addmm_(d2t, x2)
     else:
         g_view.add_(torch.matmul(d2t, x2))
+    _notify(g_view)
 
 
 class FusedArenaLinearFn(torch.autograd.Function):
@@ -79,6 +99,7 @@ class ArenaLinearFn(torch.autograd.Function):
         _acc_wgrad(ctx.g_w, d2.t(), x2)
         if ctx.g_b is not None:
             ctx.g_b.add_(d2.sum(0))
+            _notify(ctx.g_b)
         return torch.matmul(dout, w), None, None, None, None
 
 
@@ -193,3 +214,32 @@ def install_fused_projections(model: nn.Module, params_arena: torch.Tensor,
                 and mod.weight.dim() == 2):
             _wrap_single(mod, params_arena, grads_arena)
     return count
+
+
+def notify_producers(model: nn.Module):
+    """Enumerate the grad-arena ranges filled by the in-place-dW notify path
+    (fused groups + wrapped singles) as ``[(offset, numel)]``, plus the set
+    of Parameter ids those ranges cover. Params NOT covered still produce
+    grads through plain autograd (AccumulateGrad); a covered param can
+    *additionally* produce through AccumulateGrad when it is also used
+    directly (tied lm_head/embedding weight) — callers must count that as a
+    separate producer."""
+    ranges, covered = [], set()
+    for mod in model.modules():
+        for attr, names in (("_fused_qkv", ("q_proj", "k_proj", "v_proj")),
+                            ("_fused_gate_up", ("gate_proj", "up_proj")),
+                            ("_fused_kvq", ("k_proj", "v_proj", "q_proj"))):
+            got = getattr(mod, attr, None)
+            if got:
+                _w, g, _s = got
+                ranges.append((g.storage_offset(), g.numel()))
+                covered.update(id(getattr(mod, nm).weight) for nm in names)
+        af = getattr(mod, "_arena_fuse", None)
+        if af is not None:
+            _w, g_w, _b, g_b = af
+            ranges.append((g_w.storage_offset(), g_w.numel()))
+            covered.add(id(mod.weight))
+            if g_b is not None:
+                ranges.append((g_b.storage_offset(), g_b.numel()))
+                covered.add(id(mod.bias))
+    return ranges, covered

@@ -82,6 +82,13 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     world = max(env_world, 1)
+    if args.gpus != world:
+        # refuse to emit a record labeled n_gpus=K while actually running
+        # on WORLD_SIZE=W ranks (the driver launches one rank per GPU)
+        print(f"--gpus {args.gpus} != WORLD_SIZE {world}; launch with "
+              f"torch.distributed.run --nproc-per-node {args.gpus}",
+              file=sys.stderr)
+        sys.exit(2)
 
     cuda = torch.cuda.is_available()
     if cuda:
@@ -127,12 +134,12 @@ def main():
     comm = CommBackend(device)
     comm.all_reduce_avg(params)
 
-    if args.method != "ddp":
-        # (DDP keeps per-param GEMMs so its backward-overlap hooks fire)
-        from acco_amd.models.fuse import install_fused_projections
-        n_fused = install_fused_projections(model, params, grads)
-        if rank == 0:
-            print(f"# fused projection groups: {n_fused}", file=sys.stderr)
+    # fused projections run in every mode (DDP readiness tracking is
+    # element-coverage based) so ACCO vs DDP compares identical compute
+    from acco_amd.models.fuse import install_fused_projections
+    n_fused = install_fused_projections(model, params, grads)
+    if rank == 0:
+        print(f"# fused projection groups: {n_fused}", file=sys.stderr)
 
     opt = ShardedAdamW(spec, rank, device, lr=args.lr, betas=(0.9, 0.95),
                        eps=1e-8, weight_decay=0.1)

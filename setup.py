@@ -17,8 +17,12 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 
-sources = sorted(glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.cpp"))
-                 + glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.hip")))
+sources = sorted(
+    s for s in (glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.cpp"))
+                + glob.glob(os.path.join(HERE, "acco_amd", "ops", "csrc", "*.hip")))
+    # torch-hipify regenerates *_hip.hip twins at build time; compiling both
+    # the hand-written file and a stale twin double-defines every kernel
+    if not s.endswith("_hip.hip"))
 
 setup(
     name="acco_amd_hip_ops",
