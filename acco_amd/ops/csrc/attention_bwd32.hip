@@ -103,8 +103,10 @@ ACCO_DEV void stage_R(const u16* src, long long stride, u16* dst) {
 }
 
 // ------------------------------------------------------------------- dQ
+// D=64 capped at 128 VGPR: 2 co-resident 8-wave blocks per CU, matching
+// the forward kernel's occupancy choice.
 template <int D>
-__global__ __launch_bounds__(512)
+__global__ __launch_bounds__(512, D == 64 ? 4 : 2)
 void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                           const u16* __restrict__ v, const u16* __restrict__ dO,
                           const float* __restrict__ lse,
@@ -141,8 +143,11 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     dof[s] = *reinterpret_cast<const short8*>(
         dOp + (long long)lq * do_rs + s * 16 + hi * 8);
   }
-  const float lse_c = lse[(long long)bh * S + q0 + lq];
-  const float delta_c = delta[(long long)bh * S + q0 + lq];
+  // log2-domain P recompute: lse pre-scaled by log2e, delta by ·scale so
+  // the per-element work is one mul, one exp2 and one fma
+  const float lse2_c = lse[(long long)bh * S + q0 + lq] * 1.4426950408889634f;
+  const float delta_s = delta[(long long)bh * S + q0 + lq] * scale;
+  const float scale2 = scale * 1.4426950408889634f;
 
   f32x16 acc_dq[DT];
 #pragma unroll
@@ -183,16 +188,24 @@ void attn_bwd32_dq_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
       }
       __builtin_amdgcn_s_setprio(0);
 
-      // dS^T (C: col=q=lq, row=kv spread)
+      // dS^T (C: col=q=lq, row=kv spread); interior tiles (all kv ≤ all q,
+      // all within the window) skip the per-element mask predicates
       const int q_g = q0 + lq;
+      const bool need_mask = (j * KT + KT - 1 > q0) ||
+                             (window > 0 && j * KT <= q_wave_max - window);
       float ds16[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int kv_g = j * KT + m32 * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        bool valid = (kv_g <= q_g);
-        if (window > 0) valid = valid && (kv_g > q_g - window);
-        const float pval = valid ? __expf(st[r] * scale - lse_c) : 0.0f;
-        ds16[r] = pval * (dpt[r] - delta_c) * scale;
+        float x = st[r] * scale2 - lse2_c;
+        if (need_mask) {
+          const int kv_g =
+              j * KT + m32 * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          bool valid = (kv_g <= q_g);
+          if (window > 0) valid = valid && (kv_g > q_g - window);
+          x = valid ? x : -1e30f;
+        }
+        const float pval = __builtin_amdgcn_exp2f(x);
+        ds16[r] = pval * __builtin_fmaf(dpt[r], scale, -delta_s);
       }
 
       // dQ += dS·K (A = dS[row=q][k=kv] via swap; B = K^T from kT_lds)
@@ -375,9 +388,12 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
 
 #pragma unroll
     for (int m32 = 0; m32 < 2; ++m32) {
-      // per-lane lse/delta for q = qt*KT + m32*32 + lq (broadcast by row)
-      const float lse_l = lse[(long long)bh * S + qt * KT + m32 * 32 + lq];
-      const float del_l = delta[(long long)bh * S + qt * KT + m32 * 32 + lq];
+      // per-lane lse/delta for q = qt*KT + m32*32 + lq (broadcast by row),
+      // pre-scaled: lse → log2 domain, delta → ·scale (fma fold below)
+      const float lse_l =
+          lse[(long long)bh * S + qt * KT + m32 * 32 + lq] * 1.4426950408889634f;
+      const float del_l =
+          delta[(long long)bh * S + qt * KT + m32 * 32 + lq] * scale;
 
       f32x16 st, dpt;
 #pragma unroll
@@ -417,17 +433,27 @@ void attn_bwd32_dkv_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
       // st/dpt in place: separate p16/ds16 arrays cost 32 VGPRs and spill
       // the D=128 instantiation
       const int kv_g = kv0 + lq;
+      const float scale2_ = scale * 1.4426950408889634f;
+      // interior q tiles (all q ≥ all kv of this wave, all within window)
+      // skip the per-element mask predicates; (q_g < S) holds by S % KT == 0
+      const bool need_mask =
+          (qt * KT < kv0 + QW - 1) ||
+          (window > 0 && kv0 + window <= qt * KT + KT - 1);
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int rr = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const int q_g = qt * KT + m32 * 32 + rr;
         const float lse_q = __shfl(lse_l, rr, 64);
         const float del_q = __shfl(del_l, rr, 64);
-        bool valid = (kv_g <= q_g) && (q_g < S);
-        if (window > 0) valid = valid && (kv_g > q_g - window);
-        const float pval = valid ? __expf(st[r] * scale - lse_q) : 0.0f;
+        float x = st[r] * scale2_ - lse_q;
+        if (need_mask) {
+          const int q_g = qt * KT + m32 * 32 + rr;
+          bool valid = (kv_g <= q_g);
+          if (window > 0) valid = valid && (kv_g > q_g - window);
+          x = valid ? x : -1e30f;
+        }
+        const float pval = __builtin_amdgcn_exp2f(x);
         st[r] = pval;
-        dpt[r] = pval * (dpt[r] - del_q) * scale;
+        dpt[r] = pval * __builtin_fmaf(dpt[r], scale, -del_q);
       }
 
       // dV += P^T·dO ; dK += dS^T·Q   (A rows = kv via swap; B from LDS^T)

@@ -35,8 +35,11 @@ ACCO_DEV unsigned pack_bf16(float lo, float hi) {
   return (unsigned)f32_to_bf16(lo) | ((unsigned)f32_to_bf16(hi) << 16);
 }
 
+// D=64 is capped at 128 VGPR: 2 co-resident 8-wave blocks per CU (4
+// waves/SIMD) out-weigh scheduling freedom — measured faster than the
+// uncapped 159-VGPR build and than a 204-VGPR T14 double-buffered variant.
 template <int D>
-__global__ __launch_bounds__(512)
+__global__ __launch_bounds__(512, D == 64 ? 4 : 2)
 void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
                        const u16* __restrict__ v, u16* __restrict__ o,
                        float* __restrict__ lse,        // [B, H, S]
@@ -86,6 +89,8 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
   }
   const int j_hi = (qt * QT + QT - 1) / KT;
   const int q_wave_max = q0 + QW - 1;
+  const float scale2 = scale * 1.4426950408889634f;   // log2 domain
+  constexpr float THR2 = 8.0f * 1.4426950408889634f;  // defer-max (T13)
 
   for (int j = j_lo; j <= j_hi; ++j) {
     // ---- stage K (row-major copy) + V (transposed) for all 8 waves
@@ -118,8 +123,11 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
     __syncthreads();
 
-    // fully-masked tile for this wave (kv all in the future): skip compute
-    if (j * KT > q_wave_max) continue;
+    // fully-masked tile for this wave (kv all future, or all outside the
+    // local window): skip compute
+    if (j * KT > q_wave_max ||
+        (window > 0 && j * KT + KT - 1 <= q0 - window))
+      continue;
 
     // ---- S^T: st[m32] per 32-kv sub-tile (C: col=q=lq, row=kv spread)
     f32x16 st[2];
@@ -139,9 +147,13 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- mask + online softmax (per q col = lq)
+    // ---- mask + online softmax (per q col = lq), log2 domain: scores are
+    // s·scale·log2e and raw v_exp_f32 (= 2^x) replaces mul+exp per element.
+    // WAVE-level mask class: interior tiles (all kv ≤ all q, all in window)
+    // skip the per-element predicates entirely.
     const int q_g = q0 + lq;
-    const bool diag = (j * KT + KT - 1) > (qt * QT);
+    const bool need_mask = (j * KT + KT - 1 > q0) ||
+                           (window > 0 && j * KT <= q_wave_max - window);
     float p[32];
     float tmax = -1e30f;
 #pragma unroll
@@ -150,8 +162,8 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
       for (int r = 0; r < 16; ++r) {
         // row = (r&3) + 8*(r>>2) + 4*hi within the 32-kv sub-tile
         const int kv_g = j * KT + m32 * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float x = st[m32][r] * scale;
-        if (diag || window > 0) {
+        float x = st[m32][r] * scale2;
+        if (need_mask) {
           bool valid = (kv_g <= q_g);
           if (window > 0) valid = valid && (kv_g > q_g - window);
           x = valid ? x : -1e30f;
@@ -160,28 +172,36 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
         tmax = fmaxf(tmax, x);
       }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
-    const float m_new = fmaxf(m_c, tmax);
-    const float alpha = __expf(m_c - m_new);
+
+    // ---- T13 defer-max: skip the O-rescale pass (16 shfl + 32 mul) while
+    // the wave's max grows ≤ 8 natural units (P bounded by e^8, fine in
+    // f32 accumulation). Decision precedes this tile's exponentials and
+    // the previous tile's PV is complete — the safe textbook order.
+    float m_new = m_c;
+    if (!__all(tmax <= m_c + THR2)) {
+      m_new = fmaxf(m_c, tmax);
+      const float alpha = __builtin_amdgcn_exp2f(m_c - m_new);
+      l_c *= alpha;
+      float alpha_row[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        alpha_row[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+#pragma unroll
+      for (int t = 0; t < DT; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc_o[t][r] *= alpha_row[r];
+      m_c = m_new;
+    }
     float rsum = 0.0f;
 #pragma unroll
     for (int i = 0; i < 32; ++i) {
-      float e = (p[i] > -9e29f) ? __expf(p[i] - m_new) : 0.0f;
+      const float e =
+          (p[i] > -9e29f) ? __builtin_amdgcn_exp2f(p[i] - m_new) : 0.0f;
       p[i] = e;
       rsum += e;
     }
     rsum += __shfl_xor(rsum, 32, 64);
-    l_c = l_c * alpha + rsum;
-    m_c = m_new;
-
-    // ---- rescale O (row layout: q = (r&3)+8*(r>>2)+4*hi)
-    float alpha_row[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r)
-      alpha_row[r] = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
-#pragma unroll
-    for (int t = 0; t < DT; ++t)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) acc_o[t][r] *= alpha_row[r];
+    l_c += rsum;
 
     // ---- P (C layout) → A fragments in-register, then PV
     // Per 16-kv K-step kk the A fragment's 4 dwords are kv pairs
@@ -235,9 +255,12 @@ void attn_fwd32_kernel(const u16* __restrict__ q, const u16* __restrict__ k,
       Op[(long long)qrow * o_rs + t * 32 + lq] =
           f32_to_bf16(acc_o[t][r] * inv_l);
   }
+  // lse stays in NATURAL units (the backward consumes exp(s·scale − lse));
+  // the running m_c/l_c are log2-domain, so lse = ln2·(m2 + log2 l)
   if (hi == 0)
     lse[((long long)bh) * S + q0 + lq] =
-        m_c + __logf(fmaxf(l_c, 1e-30f));
+        (m_c + __builtin_amdgcn_logf(fmaxf(l_c, 1e-30f))) *
+        0.6931471805599453f;
 }
 
 }  // namespace

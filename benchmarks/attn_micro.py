@@ -24,7 +24,7 @@ def main():
     o, l = ops.hip_ext().attn_fwd(q, k, v, sc, 0)
     delta = (do.float() * o.float()).sum(-1).permute(0, 2, 1).contiguous()
 
-    iters = 10
+    iters = 20
     for _ in range(3):
         ops.hip_ext().attn_fwd(q, k, v, sc, 0)
         ops.hip_ext().attn_bwd(q, k, v, do, l, delta, sc, 0)
@@ -32,11 +32,20 @@ def main():
     t = time.time()
     for _ in range(iters):
         ops.hip_ext().attn_fwd(q, k, v, sc, 0)
+    torch.cuda.synchronize()
+    dt_f = (time.time() - t) / iters
+    t = time.time()
+    for _ in range(iters):
         ops.hip_ext().attn_bwd(q, k, v, do, l, delta, sc, 0)
     torch.cuda.synchronize()
-    dt = (time.time() - t) / iters
-    fl = 2 * 2 * B * H * S * S * Dh * 0.5 * (1 + 2.5)
-    print(f"attn fwd+bwd: {dt*1e3:.3f} ms, ~{fl/dt/1e12:.0f} TF/s combined")
+    dt_b = (time.time() - t) / iters
+    fl_f = 2 * 2 * B * H * S * S * Dh * 0.5          # causal: half the tiles
+    fl_b = fl_f * 2.5
+    print(f"attn fwd:  {dt_f*1e3:.3f} ms, ~{fl_f/dt_f/1e12:.0f} TF/s")
+    print(f"attn bwd:  {dt_b*1e3:.3f} ms, ~{fl_b/dt_b/1e12:.0f} TF/s")
+    dt = dt_f + dt_b
+    print(f"attn fwd+bwd: {dt*1e3:.3f} ms, "
+          f"~{(fl_f+fl_b)/dt/1e12:.0f} TF/s combined")
 
 
 if __name__ == "__main__":
