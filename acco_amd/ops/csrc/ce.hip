@@ -43,14 +43,19 @@ __global__ void ce_fwd_kernel(const u16* __restrict__ logits,  // [T, V]
                     bf16_to_f32(b.z), bf16_to_f32(b.w)};
       // chunk max first, one rescale of the running sum per 8 elements —
       // replaces the per-element branchy online update (8 dependent
-      // branch+rescale chains) with independent exps
-      float cm = f[0];
-#pragma unroll
-      for (int k = 1; k < 8; ++k) cm = fmaxf(cm, f[k]);
+      // branch+rescale chains) with independent exps. The 8 exps combine as
+      // a TREE: without fast-math the naive `sum += e[k]` loop is a strictly
+      // ordered 9-deep dependent fp chain per chunk; the tree cuts the
+      // cross-chunk dependency to rescale-mul + one add.
+      float cm = fmaxf(fmaxf(fmaxf(f[0], f[1]), fmaxf(f[2], f[3])),
+                       fmaxf(fmaxf(f[4], f[5]), fmaxf(f[6], f[7])));
       const float mn = fmaxf(m, cm);
-      sum *= __expf(m - mn);
+      float e[8];
 #pragma unroll
-      for (int k = 0; k < 8; ++k) sum += __expf(f[k] - mn);
+      for (int k = 0; k < 8; ++k) e[k] = __expf(f[k] - mn);
+      const float t = ((e[0] + e[1]) + (e[2] + e[3])) +
+                      ((e[4] + e[5]) + (e[6] + e[7]));
+      sum = sum * __expf(m - mn) + t;
       m = mn;
     }
     for (int tail = nv * 8 + threadIdx.x; tail < V; tail += BLOCK) {
