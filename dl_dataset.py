@@ -20,9 +20,11 @@ def main(argv=None):
     import datasets
     from transformers import AutoTokenizer
 
-    ds = datasets.load_dataset(cfg.data.path)
+    from acco_amd.data import load_raw_dataset, tokenizer_path
+
+    ds = load_raw_dataset(cfg.data)
     ds = ds["train"].train_test_split(0.05, seed=42)   # reference main.py:50
-    tokenizer = AutoTokenizer.from_pretrained(cfg.model.tokenizer)
+    tokenizer = AutoTokenizer.from_pretrained(tokenizer_path(cfg))
     tokenizer.pad_token_id = tokenizer.eos_token_id
 
     fn = make_tokenize_const_len_fn(tokenizer, "text", cfg.train.max_length)

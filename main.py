@@ -35,13 +35,14 @@ def build_datasets(cfg, model):
                                           seed=cfg.seed + 1)
                  if data.n_eval_sequences else None)
         return train, evald, None
-    # HF path (needs network or local cache): same split policy as the
-    # reference (train_test_split 0.05 seed 42, reference main.py:50)
-    import datasets
+    # HF path (hub, or a committed local corpus via kind=hf_local): same
+    # split policy as the reference (train_test_split 0.05 seed 42,
+    # reference main.py:50)
     from transformers import AutoTokenizer
-    ds = datasets.load_dataset(data.path)
+    from acco_amd.data import load_raw_dataset, tokenizer_path
+    ds = load_raw_dataset(data)
     ds = ds["train"].train_test_split(0.05, seed=42)
-    tokenizer = AutoTokenizer.from_pretrained(cfg.model.tokenizer)
+    tokenizer = AutoTokenizer.from_pretrained(tokenizer_path(cfg))
     tokenizer.pad_token_id = tokenizer.eos_token_id
     return ds["train"], ds["test"], tokenizer
 

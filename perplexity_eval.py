@@ -58,11 +58,11 @@ def main(argv=None):
                                       cfg.train.max_length,
                                       model.cfg.vocab_size, seed=cfg.seed)
     else:
-        import datasets
         from transformers import AutoTokenizer
+        from acco_amd.data import load_raw_dataset, tokenizer_path
         from acco_amd.data.packing import make_tokenize_const_len_fn
-        raw = datasets.load_dataset(cfg.data.path)["train"]
-        tok = AutoTokenizer.from_pretrained(cfg.model.tokenizer)
+        raw = load_raw_dataset(cfg.data)["train"]
+        tok = AutoTokenizer.from_pretrained(tokenizer_path(cfg))
         tok.pad_token_id = tok.eos_token_id
         ds = raw.map(make_tokenize_const_len_fn(tok, "text",
                                                 cfg.train.max_length),
