@@ -84,6 +84,13 @@ class DecoupledTrainer:
         from acco_amd.models.fuse import install_fused_projections
         install_fused_projections(model, self.params, self.grads)
 
+        # per-layer activation recompute (8B-on-288GB batch headroom)
+        if getattr(args, "activation_checkpointing", False):
+            inner = getattr(model, "model", None) or getattr(
+                model, "transformer", None)
+            if inner is not None:
+                inner.gradient_checkpointing = True
+
         # ---- observability
         out_dir = os.getcwd()
         self.scalars = ScalarLogger(os.path.join(out_dir, "scalars"),

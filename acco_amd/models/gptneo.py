@@ -112,13 +112,21 @@ class GPTNeoModel(nn.Module):
         self.wpe = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
         self.h = nn.ModuleList(GPTNeoBlock(cfg, i) for i in range(cfg.num_layers))
         self.ln_f = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
+        # per-layer activation recompute (train.activation_checkpointing)
+        self.gradient_checkpointing = False
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         S = input_ids.shape[1]
         pos = torch.arange(S, device=input_ids.device)
         x = self.wte(input_ids) + self.wpe(pos)[None]
-        for block in self.h:
-            x = block(x)
+        if (self.gradient_checkpointing and self.training
+                and torch.is_grad_enabled()):
+            from torch.utils.checkpoint import checkpoint
+            for block in self.h:
+                x = checkpoint(block, x, use_reentrant=False)
+        else:
+            for block in self.h:
+                x = block(x)
         return ops.layer_norm(x, self.ln_f.weight, self.ln_f.bias, self.ln_f.eps)
 
 

@@ -124,6 +124,10 @@ class LlamaModel(nn.Module):
         # RoPE tables are buffers recomputed lazily per device/seq-len
         # (host-precomputed trig per guide Appendix B: no on-device sinf/cosf).
         self._rope_cache: dict = {}
+        # per-layer activation recompute (train.activation_checkpointing —
+        # 8B-on-288GB batch headroom); deterministic
+        # forward (no dropout), so plain non-reentrant checkpointing is exact
+        self.gradient_checkpointing = False
 
     def _cos_sin(self, S: int, device):
         key = (S, device)
@@ -138,8 +142,14 @@ class LlamaModel(nn.Module):
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         x = self.embed_tokens(input_ids)
         cos, sin = self._cos_sin(input_ids.shape[1], x.device)
-        for layer in self.layers:
-            x = layer(x, cos, sin)
+        if (self.gradient_checkpointing and self.training
+                and torch.is_grad_enabled()):
+            from torch.utils.checkpoint import checkpoint
+            for layer in self.layers:
+                x = checkpoint(layer, x, cos, sin, use_reentrant=False)
+        else:
+            for layer in self.layers:
+                x = layer(x, cos, sin)
         return self.norm(x)
 
 

@@ -7,13 +7,19 @@ C1-C4) against a *bucket-major* flat-buffer layout:
 
 Each (bucket, rank) segment is contiguous, so every collective is a
 contiguous in-place reduce-scatter / all-gather on one bucket. Rationale
-(MI355X): each GPU has 7 point-to-point xGMI links (~153 GB/s each); a
-single monolithic ring collective is per-link bound, while nb(=8 default)
-independently launched bucket collectives let RCCL spread channels across
-links AND let the sharded-AdamW kernel for bucket j start as soon as bucket
-j's reduce-scatter lands — pipelining optimizer compute under the remaining
-buckets' communication (the reference issues one monolithic collective pair,
-trainer_decoupled.py:88-112).
+(MI355X): collectives on ONE communicator execute in issue order — bucket
+count does NOT add cross-collective link parallelism (RCCL spreads a
+single collective over its channels/links by itself, controlled by
+NCCL_MIN_NCHANNELS). What bucketing buys is the PIPELINE: the sharded
+fused-AdamW for bucket j launches as soon as bucket j's reduce-scatter
+lands and bucket j's all-gather launches right after its AdamW, so
+optimizer compute and the remaining buckets' communication overlap, and
+the first all-gather bytes are on the xGMI links long before the last
+reduce-scatter finishes (the reference issues one monolithic pair with the
+full optimizer step serialized between them, trainer_decoupled.py:88-112).
+Per-bucket size also stays large enough (~34 MB at llama-1b world=8,
+nb=8) to amortize per-collective launch cost on the 7×153 GB/s
+point-to-point fabric; `comm_buckets` is the tuning knob.
 
 The flat order of buffer[:N] is the model's parameter order, so
 params ↔ buffer copies stay a single contiguous cast-copy; ownership of a

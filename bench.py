@@ -65,6 +65,8 @@ def parse_args():
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--buckets", type=int, default=8)
     p.add_argument("--lr", type=float, default=6e-4)
+    p.add_argument("--ckpt", action="store_true",
+                   help="per-layer activation recompute (8B batch headroom)")
     return p.parse_args()
 
 
@@ -140,6 +142,8 @@ def main():
     n_fused = install_fused_projections(model, params, grads)
     if rank == 0:
         print(f"# fused projection groups: {n_fused}", file=sys.stderr)
+    if args.ckpt:
+        model.model.gradient_checkpointing = True
 
     opt = ShardedAdamW(spec, rank, device, lr=args.lr, betas=(0.9, 0.95),
                        eps=1e-8, weight_decay=0.1)
