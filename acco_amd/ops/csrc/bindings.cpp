@@ -262,12 +262,13 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
   const int S = (int)logits.size(1), V = (int)logits.size(2);
   const long long T = Bn * S;
   auto lse = at::empty({T}, logits.options().dtype(at::kFloat));
-  auto acc = at::zeros({2}, logits.options().dtype(at::kFloat));
+  // 128 accumulator shards (atomic-contention fix in ce.hip) reduced here
+  auto accs = at::zeros({128, 2}, logits.options().dtype(at::kFloat));
   acco_ce_fwd(logits.data_ptr(),
               reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
-              lse.data_ptr<float>(), acc.data_ptr<float>(), T, S, V,
+              lse.data_ptr<float>(), accs.data_ptr<float>(), T, S, V,
               cur_stream());
-  return {acc, lse};
+  return {accs.sum(0), lse};
 }
 
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,

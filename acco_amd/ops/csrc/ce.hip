@@ -85,8 +85,13 @@ __global__ void ce_fwd_kernel(const u16* __restrict__ logits,  // [T, V]
       const float l = M + __logf(Ssum);
       lse[t] = l;
       const float gold = bf16_to_f32(row[label]);
-      atomicAdd(acc, l - gold);
-      atomicAdd(acc + 1, 1.0f);
+      // SHARDED accumulator: 16k per-row atomicAdds on one L2 cell
+      // serialize at ~88/us and were the kernel's real bound (PMC: 87%
+      // WAIT_ANY, loads already dwordx4); 128 shards cut the contention
+      // 128x and the wrapper reduces [128,2] with one tiny sum
+      float* a = acc + 2 * (blockIdx.x & 127);
+      atomicAdd(a, l - gold);
+      atomicAdd(a + 1, 1.0f);
     }
     __syncthreads();
   }

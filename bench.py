@@ -262,6 +262,9 @@ def main():
                                 if args.method != "ddp" else f"ddp{world}"),
                 "comm_buckets": spec.nb,
                 "params": n_live,
+                "ckpt": bool(args.ckpt),
+                "peak_mem_gb": (round(torch.cuda.max_memory_allocated()
+                                      / 2**30, 1) if cuda else None),
             },
         }))
 
