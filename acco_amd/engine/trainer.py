@@ -239,11 +239,12 @@ class DecoupledTrainer:
         with self._autocast():
             if ls != 0.0:
                 # label-smoothed path (reference compute_loss :262-282 with
-                # LabelSmoother): loss computed outside the model head
-                from acco_amd.ops import torch_ref
+                # LabelSmoother): loss computed outside the model head —
+                # fused into the CE HIP kernel on GPU (K9)
+                from acco_amd import ops as _ops
                 labels = inputs.get("labels", inputs["input_ids"])
                 logits = self.model(inputs["input_ids"])[0]
-                loss = torch_ref.label_smoothed_causal_lm_loss(
+                loss = _ops.label_smoothed_causal_lm_loss(
                     logits, labels, ls) / self.loss_div
             else:
                 if "labels" in inputs:

@@ -20,12 +20,35 @@ _EXT = None
 _EXT_ERR: Optional[str] = None
 
 
+def _check_fresh(mod) -> None:
+    """Warn if the in-tree .so predates any kernel source (a stale binary
+    silently satisfies the import and defeats the fail-loud policy —
+    round-1 advisor finding)."""
+    import glob
+    import warnings
+    so = getattr(mod, "__file__", None)
+    if not so or not os.path.exists(so):
+        return
+    so_mtime = os.path.getmtime(so)
+    csrc = os.path.join(os.path.dirname(so), "ops", "csrc")
+    stale = [os.path.basename(f)
+             for f in glob.glob(os.path.join(csrc, "*.hip"))
+             + glob.glob(os.path.join(csrc, "*.cpp"))
+             if not f.endswith("_hip.hip") and os.path.getmtime(f) > so_mtime]
+    if stale:
+        warnings.warn(
+            f"acco_amd._hip_ops is OLDER than kernel sources {stale}; "
+            "rebuild with `python setup.py build_ext --inplace`",
+            RuntimeWarning, stacklevel=3)
+
+
 def _load_ext():
     global _EXT, _EXT_ERR
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
     try:
         from acco_amd import _hip_ops  # built by setup.py build_ext --inplace
+        _check_fresh(_hip_ops)
         _EXT = _hip_ops
     except ImportError as e:  # remember why, for the loud failure below
         _EXT_ERR = str(e)
@@ -127,6 +150,18 @@ def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
         return torch_ref.causal_lm_loss(logits, labels)
     from acco_amd.ops.autograd import CausalLMLossFn
     return CausalLMLossFn.apply(logits, labels)
+
+
+def label_smoothed_causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor,
+                                  epsilon: float) -> torch.Tensor:
+    """HF-LabelSmoother-equivalent shifted loss fused into the CE kernel
+    (reference utils/trainer_utils.py:862-902)."""
+    if epsilon == 0.0:
+        return causal_lm_loss(logits, labels)
+    if _use_ref(logits, "ce_fwd"):
+        return torch_ref.label_smoothed_causal_lm_loss(logits, labels, epsilon)
+    from acco_amd.ops.autograd import CausalLMLossFn
+    return CausalLMLossFn.apply(logits, labels, epsilon)
 
 
 # ------------------------------------------------------------- trainer ops

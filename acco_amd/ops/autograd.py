@@ -114,14 +114,17 @@ class RoPEFn(torch.autograd.Function):
 
 
 class CausalLMLossFn(torch.autograd.Function):
-    """Fused shifted CE: logits [B,S,V] bf16, labels [B,S] int64."""
+    """Fused shifted CE: logits [B,S,V] bf16, labels [B,S] int64.
+    epsilon != 0 adds HF-LabelSmoother semantics inside the same kernel
+    pass."""
 
     @staticmethod
-    def forward(ctx, logits, labels):
+    def forward(ctx, logits, labels, epsilon=0.0):
         logits = logits.contiguous()
         labels = labels.contiguous()
-        acc, lse = ops.hip_ext().ce_fwd(logits, labels)
+        acc, lse = ops.hip_ext().ce_fwd(logits, labels, float(epsilon))
         ctx.save_for_backward(logits, labels, lse, acc)
+        ctx.epsilon = float(epsilon)
         return acc[0] / acc[1].clamp(min=1.0)
 
     @staticmethod
@@ -132,10 +135,11 @@ class CausalLMLossFn(torch.autograd.Function):
             # read the upstream grad on-device — float(grad_out) would be a
             # D2H sync stalling the backward launch pipeline every micro
             dlogits = ext.ce_bwd_dev(logits, labels, lse, acc,
-                                     grad_out.reshape(1).float())
+                                     grad_out.reshape(1).float(), ctx.epsilon)
         else:
-            dlogits = ext.ce_bwd(logits, labels, lse, acc, float(grad_out))
-        return dlogits, None
+            dlogits = ext.ce_bwd(logits, labels, lse, acc, float(grad_out),
+                                 ctx.epsilon)
+        return dlogits, None, None
 
 
 class AttnQKVPackedFn(torch.autograd.Function):

@@ -34,10 +34,10 @@ extern "C" void acco_rope(const void*, void*, const float*, const float*,
                           long long, int, int, int, bool, long long,
                           hipStream_t);
 extern "C" void acco_ce_fwd(const void*, const long long*, float*, float*,
-                            long long, int, int, hipStream_t);
+                            long long, int, int, float, hipStream_t);
 extern "C" void acco_ce_bwd(const void*, const long long*, const float*,
                             void*, const float*, float, const float*,
-                            long long, int, int, hipStream_t);
+                            long long, int, int, float, hipStream_t);
 extern "C" void acco_gemm_nt(const void*, const void*, void*, int, int,
                              int, hipStream_t);
 extern "C" void acco_attn_fwd(const void*, const void*, const void*, void*,
@@ -254,7 +254,8 @@ at::Tensor rope_fwd(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
 }
 
 // ---- fused shifted causal-LM CE
-std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
+                               double epsilon) {
   CHECK_BF16_CONTIG(logits);
   TORCH_CHECK(logits.dim() == 3 && labels.dim() == 2);
   TORCH_CHECK(labels.scalar_type() == at::kLong && labels.is_contiguous());
@@ -267,12 +268,12 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels) {
   acco_ce_fwd(logits.data_ptr(),
               reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
               lse.data_ptr<float>(), accs.data_ptr<float>(), T, S, V,
-              cur_stream());
+              (float)epsilon, cur_stream());
   return {accs.sum(0), lse};
 }
 
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
-                  at::Tensor acc, double dloss) {
+                  at::Tensor acc, double dloss, double epsilon) {
   CHECK_BF16_CONTIG(logits);
   const long long Bn = logits.size(0);
   const int S = (int)logits.size(1), V = (int)logits.size(2);
@@ -282,13 +283,13 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor lse,
               reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
               lse.data_ptr<float>(), dlogits.data_ptr(),
               acc.data_ptr<float>(), (float)dloss, nullptr, T, S, V,
-              cur_stream());
+              (float)epsilon, cur_stream());
   return dlogits;
 }
 
 // dloss read from a 1-elem fp32 device tensor: no D2H sync in backward
 at::Tensor ce_bwd_dev(at::Tensor logits, at::Tensor labels, at::Tensor lse,
-                      at::Tensor acc, at::Tensor dloss_dev) {
+                      at::Tensor acc, at::Tensor dloss_dev, double epsilon) {
   CHECK_BF16_CONTIG(logits);
   TORCH_CHECK(dloss_dev.is_cuda() && dloss_dev.scalar_type() == at::kFloat &&
               dloss_dev.numel() == 1);
@@ -300,7 +301,7 @@ at::Tensor ce_bwd_dev(at::Tensor logits, at::Tensor labels, at::Tensor lse,
               reinterpret_cast<const long long*>(labels.data_ptr<int64_t>()),
               lse.data_ptr<float>(), dlogits.data_ptr(),
               acc.data_ptr<float>(), 0.0f, dloss_dev.data_ptr<float>(), T, S,
-              V, cur_stream());
+              V, (float)epsilon, cur_stream());
   return dlogits;
 }
 
@@ -447,9 +448,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("rope_fwd", &rope_fwd);
-  m.def("ce_fwd", &ce_fwd);
-  m.def("ce_bwd", &ce_bwd);
-  m.def("ce_bwd_dev", &ce_bwd_dev);
+  m.def("ce_fwd", &ce_fwd, py::arg("logits"), py::arg("labels"),
+        py::arg("epsilon") = 0.0);
+  m.def("ce_bwd", &ce_bwd, py::arg("logits"), py::arg("labels"),
+        py::arg("lse"), py::arg("acc"), py::arg("dloss"),
+        py::arg("epsilon") = 0.0);
+  m.def("ce_bwd_dev", &ce_bwd_dev, py::arg("logits"), py::arg("labels"),
+        py::arg("lse"), py::arg("acc"), py::arg("dloss_dev"),
+        py::arg("epsilon") = 0.0);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);

@@ -309,38 +309,59 @@ void layernorm_bwd_wr_kernel(const u16* __restrict__ dy,
     for (int kk = 0; kk < VEC; ++kk) { dwacc[j][kk] = 0.f; dbacc[j][kk] = 0.f; }
   }
 
-  for (long long row = (long long)blockIdx.x * 4 + g; row < R;
-       row += (long long)gridDim.x * 4) {
-    const u16* dyr = dy + row * D;
-    const u16* xr = x + row * D;
-    u16* dxr = dx + row * D;
-    const float mean = mean_in[row];
-    const float r = rstd_in[row];
+  // 1-deep software pipeline over rows: the next row's x/dy uint4 loads
+  // (and its mean/rstd) issue before the current row's reductions, hiding
+  // the ~900-cycle HBM latency under the shfl-reduce + epilogue (PMC
+  // showed the plain loop 55% parked at waits)
+  const long long stride = (long long)gridDim.x * 4;
+  long long row = (long long)blockIdx.x * 4 + g;
+  uint4 xv[CH], dv[CH];
+  float mean = 0.f, r = 0.f;
+  auto issue = [&](long long rw) {
+#pragma unroll
+    for (int j = 0; j < CH; ++j)
+      if (act[j]) {
+        xv[j] = *reinterpret_cast<const uint4*>(x + rw * D + cid[j] * VEC);
+        dv[j] = *reinterpret_cast<const uint4*>(dy + rw * D + cid[j] * VEC);
+      }
+    mean = mean_in[rw];
+    r = rstd_in[rw];
+  };
+  auto unpack8 = [](const uint4& v, float* f) {
+    const u16* u = reinterpret_cast<const u16*>(&v);
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) f[kk] = bf16_to_f32(u[kk]);
+  };
+  if (row < R) issue(row);
+  for (; row < R; row += stride) {
     float xh[CH][VEC], ds[CH][VEC];
+    const float mean_c = mean, r_c = r;
     float s1 = 0.0f, s2 = 0.0f;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
       if (act[j]) {
         float xs[VEC];
-        load8(xr + cid[j] * VEC, xs);
-        load8(dyr + cid[j] * VEC, ds[j]);
+        unpack8(xv[j], xs);
+        unpack8(dv[j], ds[j]);
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) {
-          xh[j][kk] = (xs[kk] - mean) * r;
+          xh[j][kk] = (xs[kk] - mean_c) * r_c;
           const float dyw = ds[j][kk] * wf[j][kk];
           s1 += dyw;
           s2 += dyw * xh[j][kk];
         }
       }
+    if (row + stride < R) issue(row + stride);   // in flight under reduce
     s1 = wave_reduce_sum(s1) / (float)D;
     s2 = wave_reduce_sum(s2) / (float)D;
+    u16* dxr = dx + row * D;
 #pragma unroll
     for (int j = 0; j < CH; ++j)
       if (act[j]) {
         float o[VEC];
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) {
-          o[kk] = r * (ds[j][kk] * wf[j][kk] - s1 - xh[j][kk] * s2);
+          o[kk] = r_c * (ds[j][kk] * wf[j][kk] - s1 - xh[j][kk] * s2);
           dwacc[j][kk] += ds[j][kk] * xh[j][kk];
           dbacc[j][kk] += ds[j][kk];
         }

@@ -127,6 +127,26 @@ def test_ce_fwd_bwd():
     assert _close_bf16(logits.grad, l32.grad, atol=1e-3, rtol=5e-2)
 
 
+def test_ce_label_smoothed_fwd_bwd():
+    """K9: label smoothing fused into the CE kernel vs the HF-parity
+    torch reference (forward value + full dlogits)."""
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    eps = 0.1
+    B, S, V = 2, 33, 1031          # odd V exercises the tail path
+    logits = (torch.randn(B, S, V, device="cuda") * 2).bfloat16().requires_grad_(True)
+    labels = torch.randint(0, V, (B, S), device="cuda")
+    labels[0, 5] = -100
+    loss = ops.label_smoothed_causal_lm_loss(logits, labels, eps)
+    ref = torch_ref.label_smoothed_causal_lm_loss(
+        logits.detach().float(), labels, eps)
+    assert torch.allclose(loss, ref, atol=2e-3, rtol=2e-3), (loss, ref)
+    loss.backward()
+    l32 = logits.detach().float().requires_grad_(True)
+    torch_ref.label_smoothed_causal_lm_loss(l32, labels, eps).backward()
+    assert _close_bf16(logits.grad, l32.grad, atol=1e-3, rtol=5e-2)
+
+
 def test_model_ops_route_to_hip():
     """The live Llama/GPT-Neo blocks must be running HIP kernels, not ATen."""
     from acco_amd import ops
