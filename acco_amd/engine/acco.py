@@ -59,6 +59,7 @@ class AccoEngine:
                  forward_backward: Callable[[Dict], torch.Tensor],
                  next_batch: Callable[[], Dict],
                  n_grad_accumulation: int = 1,
+                 grad_reduce_dtype: Optional[str] = None,
                  log=None):
         self.params = params_arena
         self.grads = grads_arena
@@ -76,6 +77,16 @@ class AccoEngine:
 
         # the communication buffer (reference com_buffer, :244-269)
         self.com_buffer = torch.zeros(spec.total, dtype=params_arena.dtype,
+                                      device=device)
+        # fp32-accumulate reduce (train.grad_reduce_dtype='fp32'): the com
+        # round runs on an fp32 shadow — cast in, reduce-scatter + AdamW +
+        # all-gather in fp32, cast back — gating the bf16 8-rank summation
+        # error the reference accepts (trainer_decoupled.py:88-93; SURVEY.md
+        # §7 "bf16 grad reduce-scatter numerics"). 2× collective bytes.
+        self._com32: Optional[torch.Tensor] = None
+        if (grad_reduce_dtype == "fp32"
+                and self.com_buffer.dtype == torch.bfloat16):
+            self._com32 = torch.zeros(spec.total, dtype=torch.float32,
                                       device=device)
         int_t = torch.int32
         self.count_grad_local = torch.zeros(1, dtype=int_t, device=device)
@@ -167,7 +178,11 @@ class AccoEngine:
     @torch.no_grad()
     def _communication_round(self, commit: bool, advance_sched: bool) -> int:
         work_count = self.comm.all_reduce_sum_async(self.count_grad_this_round)
-        rs = [self.comm.reduce_scatter_bucket_async(self.com_buffer, self.spec,
+        buf = self.com_buffer
+        if self._com32 is not None:
+            self._com32.copy_(self.com_buffer)     # grads bf16 → fp32
+            buf = self._com32
+        rs = [self.comm.reduce_scatter_bucket_async(buf, self.spec,
                                                     j, self.rank)
               for j in range(self.spec.nb)]
         work_count.wait()
@@ -177,12 +192,14 @@ class AccoEngine:
         ag = []
         for j in range(self.spec.nb):
             rs[j].wait()
-            self.opt.step_bucket(j, self.com_buffer, grad_scale=inv_count,
+            self.opt.step_bucket(j, buf, grad_scale=inv_count,
                                  commit=commit, lr=lr)
-            ag.append(self.comm.all_gather_bucket_async(self.com_buffer,
-                                                        self.spec, j, self.rank))
+            ag.append(self.comm.all_gather_bucket_async(buf, self.spec, j,
+                                                        self.rank))
         for w in ag:
             w.wait()
+        if buf is not self.com_buffer:
+            self.com_buffer.copy_(buf)             # new params fp32 → bf16
         self.opt.finish_round(commit)
         n_global = int(self.count_grad_this_round.item())
         if advance_sched:

@@ -131,6 +131,7 @@ class DecoupledTrainer:
                 forward_backward=self.forward_backward,
                 next_batch=self.load_next_batch,
                 n_grad_accumulation=int(args.n_grad_accumulation),
+                grad_reduce_dtype=getattr(args, "grad_reduce_dtype", None),
                 log=self.log)
             # seed fp32 master from current (averaged) params
             self.opt.init_master_from_buffer(self.params)
