@@ -96,12 +96,16 @@ class GPTNeoBlock(nn.Module):
         self.ln_2 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_epsilon)
         self.mlp = GPTNeoMLP(cfg)
 
-    def forward(self, x):
-        x = x + self.attn(ops.layer_norm(x, self.ln_1.weight, self.ln_1.bias,
-                                         self.ln_1.eps))
-        x = x + self.mlp(ops.layer_norm(x, self.ln_2.weight, self.ln_2.bias,
-                                        self.ln_2.eps))
-        return x
+    def forward(self, pending, residual):
+        """(pending, residual) form — every residual add fused into the
+        following LayerNorm kernel (see LlamaDecoderLayer); returns the
+        un-added MLP output + running residual, ln_f completes the last add."""
+        h1, residual = ops.add_layer_norm(pending, residual, self.ln_1.weight,
+                                          self.ln_1.bias, self.ln_1.eps)
+        a = self.attn(h1)
+        h2, residual = ops.add_layer_norm(a, residual, self.ln_2.weight,
+                                          self.ln_2.bias, self.ln_2.eps)
+        return self.mlp(h2), residual
 
 
 class GPTNeoModel(nn.Module):
@@ -118,16 +122,20 @@ class GPTNeoModel(nn.Module):
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         S = input_ids.shape[1]
         pos = torch.arange(S, device=input_ids.device)
-        x = self.wte(input_ids) + self.wpe(pos)[None]
+        pending = self.wte(input_ids) + self.wpe(pos)[None]
+        residual = None
         if (self.gradient_checkpointing and self.training
                 and torch.is_grad_enabled()):
             from torch.utils.checkpoint import checkpoint
             for block in self.h:
-                x = checkpoint(block, x, use_reentrant=False)
+                pending, residual = checkpoint(block, pending, residual,
+                                               use_reentrant=False)
         else:
             for block in self.h:
-                x = block(x)
-        return ops.layer_norm(x, self.ln_f.weight, self.ln_f.bias, self.ln_f.eps)
+                pending, residual = block(pending, residual)
+        y, _ = ops.add_layer_norm(pending, residual, self.ln_f.weight,
+                                  self.ln_f.bias, self.ln_f.eps)
+        return y
 
 
 class GPTNeoForCausalLM(nn.Module):

@@ -101,6 +101,14 @@ class LlamaMLP(nn.Module):
 
 
 class LlamaDecoderLayer(nn.Module):
+    """Pre-norm decoder layer in (pending, residual) form: every residual
+    add is FUSED into the following norm kernel (ops.add_rms_norm returns
+    both the normed value and the sum). The layer receives the previous
+    layer's un-added MLP output (`pending`) plus the running residual, and
+    returns its own un-added MLP output — the model's final norm completes
+    the last add. Equivalent math to the classic
+    x = x + attn(norm1(x)); x = x + mlp(norm2(x)) chain."""
+
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.self_attn = LlamaAttention(cfg)
@@ -108,10 +116,15 @@ class LlamaDecoderLayer(nn.Module):
         self.input_layernorm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
         self.post_attention_layernorm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
 
-    def forward(self, x, cos, sin):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
-        x = x + self.mlp(self.post_attention_layernorm(x))
-        return x
+    def forward(self, pending, residual, cos, sin):
+        n1 = self.input_layernorm
+        h1, residual = ops.add_rms_norm(pending, residual, n1.weight,
+                                        n1.variance_epsilon)
+        a = self.self_attn(h1, cos, sin)
+        n2 = self.post_attention_layernorm
+        h2, residual = ops.add_rms_norm(a, residual, n2.weight,
+                                        n2.variance_epsilon)
+        return self.mlp(h2), residual
 
 
 class LlamaModel(nn.Module):
@@ -140,17 +153,22 @@ class LlamaModel(nn.Module):
         return hit
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
-        x = self.embed_tokens(input_ids)
-        cos, sin = self._cos_sin(input_ids.shape[1], x.device)
+        pending = self.embed_tokens(input_ids)
+        residual = None
+        cos, sin = self._cos_sin(input_ids.shape[1], pending.device)
         if (self.gradient_checkpointing and self.training
                 and torch.is_grad_enabled()):
             from torch.utils.checkpoint import checkpoint
             for layer in self.layers:
-                x = checkpoint(layer, x, cos, sin, use_reentrant=False)
+                pending, residual = checkpoint(layer, pending, residual,
+                                               cos, sin, use_reentrant=False)
         else:
             for layer in self.layers:
-                x = layer(x, cos, sin)
-        return self.norm(x)
+                pending, residual = layer(pending, residual, cos, sin)
+        # final norm completes the last residual add in the same kernel
+        y, _ = ops.add_rms_norm(pending, residual, self.norm.weight,
+                                self.norm.variance_epsilon)
+        return y
 
 
 class LlamaForCausalLM(nn.Module):

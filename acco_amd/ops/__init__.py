@@ -109,6 +109,31 @@ def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
     return LayerNormFn.apply(x, weight, bias, eps)
 
 
+def add_rms_norm(x: torch.Tensor, res: Optional[torch.Tensor],
+                 weight: torch.Tensor, eps: float):
+    """(rmsnorm(x+res)·w, x+res) in ONE kernel pass (residual-add fusion);
+    res=None degenerates to a plain norm with s = x."""
+    if res is None:
+        return rms_norm(x, weight, eps), x
+    if _use_ref(x, "add_rmsnorm_fwd"):
+        s = x + res
+        return torch_ref.rms_norm(s, weight, eps), s
+    from acco_amd.ops.autograd import AddRMSNormFn
+    return AddRMSNormFn.apply(x, res, weight, eps)
+
+
+def add_layer_norm(x: torch.Tensor, res: Optional[torch.Tensor],
+                   weight: torch.Tensor, bias: torch.Tensor, eps: float):
+    """(layernorm(x+res)·w+b, x+res) in ONE kernel pass; res=None → plain."""
+    if res is None:
+        return layer_norm(x, weight, bias, eps), x
+    if _use_ref(x, "add_layernorm_fwd"):
+        s = x + res
+        return torch_ref.layer_norm(s, weight, bias, eps), s
+    from acco_amd.ops.autograd import AddLayerNormFn
+    return AddLayerNormFn.apply(x, res, weight, bias, eps)
+
+
 def gelu_new(x: torch.Tensor) -> torch.Tensor:
     if _use_ref(x, "gelu_fwd"):
         return torch_ref.gelu_new(x)

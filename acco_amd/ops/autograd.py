@@ -68,9 +68,34 @@ class RMSNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, rstd = ctx.saved_tensors
+        none = torch.empty(0, device=x.device, dtype=x.dtype)
         dx, dw = ops.hip_ext().rmsnorm_bwd(dy.contiguous(), x,
-                                           weight.contiguous(), rstd)
+                                           weight.contiguous(), rstd, none)
         return dx, dw.to(weight.dtype), None
+
+
+class AddRMSNormFn(torch.autograd.Function):
+    """Fused residual-add + RMSNorm: (y, s) = (rmsnorm(x+res)·w, bf16(x+res)).
+    The backward folds the residual branch's grad (ds) into the norm-bwd
+    kernel's dx write — the eager add/accumulate kernels at every residual
+    site disappear (guide: fuse elementwise work into the producing kernel)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, eps):
+        y, s, rstd = ops.hip_ext().add_rmsnorm_fwd(
+            x.contiguous(), res.contiguous(), weight.contiguous(), eps)
+        ctx.save_for_backward(s, weight, rstd)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, weight, rstd = ctx.saved_tensors
+        dadd = (ds.contiguous() if ds is not None
+                else torch.empty(0, device=s.device, dtype=s.dtype))
+        dx, dw = ops.hip_ext().rmsnorm_bwd(dy.contiguous(), s,
+                                           weight.contiguous(), rstd, dadd)
+        # d(x) == d(res): both inputs feed the same sum
+        return dx, dx, dw.to(weight.dtype), None
 
 
 class LayerNormFn(torch.autograd.Function):
@@ -85,10 +110,33 @@ class LayerNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, mean, rstd = ctx.saved_tensors
+        none = torch.empty(0, device=x.device, dtype=x.dtype)
         dx, dw, db = ops.hip_ext().layernorm_bwd(dy.contiguous(), x,
                                                  weight.contiguous(), mean,
-                                                 rstd)
+                                                 rstd, none)
         return dx, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+class AddLayerNormFn(torch.autograd.Function):
+    """Fused residual-add + LayerNorm (GPT-Neo blocks); see AddRMSNormFn."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps):
+        y, s, mean, rstd = ops.hip_ext().add_layernorm_fwd(
+            x.contiguous(), res.contiguous(), weight.contiguous(),
+            bias.contiguous(), eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, weight, mean, rstd = ctx.saved_tensors
+        dadd = (ds.contiguous() if ds is not None
+                else torch.empty(0, device=s.device, dtype=s.dtype))
+        dx, dw, db = ops.hip_ext().layernorm_bwd(dy.contiguous(), s,
+                                                 weight.contiguous(), mean,
+                                                 rstd, dadd)
+        return dx, dx, dw.to(weight.dtype), db.to(weight.dtype), None
 
 
 class RoPEFn(torch.autograd.Function):

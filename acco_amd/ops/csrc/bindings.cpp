@@ -19,17 +19,19 @@ extern "C" void acco_gelu_fwd(const void*, void*, long long, hipStream_t);
 extern "C" void acco_gelu_bwd(const void*, const void*, void*, long long,
                               hipStream_t);
 extern "C" void acco_rmsnorm_fwd(const void*, const void*, void*, void*,
+                                 const void*, void*,
                                  long long, int, float, hipStream_t);
 extern "C" void acco_rmsnorm_bwd(const void*, const void*, const void*,
-                                 const void*, void*, void*, long long, int,
-                                 hipStream_t);
+                                 const void*, void*, void*, const void*,
+                                 long long, int, hipStream_t);
 extern "C" int acco_norm_bwd_grid(long long R, int D);
 extern "C" void acco_layernorm_fwd(const void*, const void*, const void*,
-                                   void*, void*, void*, long long, int, float,
-                                   hipStream_t);
+                                   void*, void*, void*, const void*, void*,
+                                   long long, int, float, hipStream_t);
 extern "C" void acco_layernorm_bwd(const void*, const void*, const void*,
                                    const void*, const void*, void*, void*,
-                                   void*, long long, int, hipStream_t);
+                                   void*, const void*, long long, int,
+                                   hipStream_t);
 extern "C" void acco_rope(const void*, void*, const float*, const float*,
                           long long, int, int, int, bool, long long,
                           hipStream_t);
@@ -185,12 +187,31 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
   auto y = at::empty_like(x);
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
   acco_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), rstd.data_ptr(),
-                   R, D, (float)eps, cur_stream());
+                   nullptr, nullptr, R, D, (float)eps, cur_stream());
   return {y, rstd};
 }
 
+// fused residual add + RMSNorm: s = bf16(x + res), y = rmsnorm(s)·w
+std::vector<at::Tensor> add_rmsnorm_fwd(at::Tensor x, at::Tensor res,
+                                        at::Tensor w, double eps) {
+  CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(res); CHECK_BF16_CONTIG(w);
+  const int D = (int)w.numel();
+  TORCH_CHECK(D % 8 == 0 && D <= 16384 && x.numel() % D == 0);
+  TORCH_CHECK(res.numel() == x.numel());
+  const long long R = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto sum_out = at::empty_like(x);
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  acco_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), rstd.data_ptr(),
+                   res.data_ptr(), sum_out.data_ptr(), R, D, (float)eps,
+                   cur_stream());
+  return {y, sum_out, rstd};
+}
+
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
-                                    at::Tensor rstd) {
+                                    at::Tensor rstd,
+                                    c10::optional<at::Tensor> dadd_opt) {
+  at::Tensor dadd = dadd_opt.value_or(at::Tensor());
   CHECK_BF16_CONTIG(dy); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w);
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
@@ -198,7 +219,8 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int grid = acco_norm_bwd_grid(R, D);
   auto dw_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_rmsnorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
-                   rstd.data_ptr(), dx.data_ptr(), dw_part.data_ptr(), R, D,
+                   rstd.data_ptr(), dx.data_ptr(), dw_part.data_ptr(),
+                   (dadd.defined() && dadd.numel() > 0) ? dadd.data_ptr() : nullptr, R, D,
                    cur_stream());
   return {dx, dw_part.sum(0)};
 }
@@ -214,14 +236,36 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b,
   auto mean = at::empty({R}, x.options().dtype(at::kFloat));
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
   acco_layernorm_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
-                     mean.data_ptr(), rstd.data_ptr(), R, D, (float)eps,
-                     cur_stream());
+                     mean.data_ptr(), rstd.data_ptr(), nullptr, nullptr, R,
+                     D, (float)eps, cur_stream());
   return {y, mean, rstd};
+}
+
+// fused residual add + LayerNorm: s = bf16(x + res), y = ln(s)·w + b
+std::vector<at::Tensor> add_layernorm_fwd(at::Tensor x, at::Tensor res,
+                                          at::Tensor w, at::Tensor b,
+                                          double eps) {
+  CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(res);
+  CHECK_BF16_CONTIG(w); CHECK_BF16_CONTIG(b);
+  const int D = (int)w.numel();
+  TORCH_CHECK(D % 8 == 0 && D <= 16384 && x.numel() % D == 0);
+  TORCH_CHECK(res.numel() == x.numel());
+  const long long R = x.numel() / D;
+  auto y = at::empty_like(x);
+  auto sum_out = at::empty_like(x);
+  auto mean = at::empty({R}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  acco_layernorm_fwd(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+                     mean.data_ptr(), rstd.data_ptr(), res.data_ptr(),
+                     sum_out.data_ptr(), R, D, (float)eps, cur_stream());
+  return {y, sum_out, mean, rstd};
 }
 
 std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
-                                      at::Tensor rstd) {
+                                      at::Tensor rstd,
+                                      c10::optional<at::Tensor> dadd_opt) {
+  at::Tensor dadd = dadd_opt.value_or(at::Tensor());
   CHECK_BF16_CONTIG(dy); CHECK_BF16_CONTIG(x); CHECK_BF16_CONTIG(w);
   const int D = (int)w.numel();
   const long long R = x.numel() / D;
@@ -231,7 +275,8 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   auto db_part = at::empty({grid, D}, x.options().dtype(at::kFloat));
   acco_layernorm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
-                     dw_part.data_ptr(), db_part.data_ptr(), R, D,
+                     dw_part.data_ptr(), db_part.data_ptr(),
+                     (dadd.defined() && dadd.numel() > 0) ? dadd.data_ptr() : nullptr, R, D,
                      cur_stream());
   return {dx, dw_part.sum(0), db_part.sum(0)};
 }
@@ -444,9 +489,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
-  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("w"), py::arg("rstd"), py::arg("dadd") = py::none());
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   m.def("layernorm_fwd", &layernorm_fwd);
-  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("layernorm_bwd", &layernorm_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("w"), py::arg("mean"), py::arg("rstd"),
+        py::arg("dadd") = py::none());
+  m.def("add_layernorm_fwd", &add_layernorm_fwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("ce_fwd", &ce_fwd, py::arg("logits"), py::arg("labels"),
         py::arg("epsilon") = 0.0);

@@ -60,10 +60,15 @@ ACCO_DEV void store8(u16* p, const float* f) {
 }
 
 // ------------------------------------------------------------ RMSNorm fwd
+// res != nullptr fuses the residual add: s = bf16(x + res) is written to
+// sum_out and the statistics/normalization run on s — one kernel replaces
+// the eager add + norm pair (saves a full read+write pass of the hidden
+// state per fusion site; the bf16 rounding of s matches the eager add).
 template <int CH, int G>
 __global__ __launch_bounds__(BLOCK)
 void rmsnorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
                         u16* __restrict__ y, float* __restrict__ rstd,
+                        const u16* __restrict__ res, u16* __restrict__ sum_out,
                         long long R, int D, float eps) {
   constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
@@ -97,6 +102,14 @@ void rmsnorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
     for (int j = 0; j < CH; ++j)
       if (live && act[j]) {
         load8(xr + cid[j] * VEC, xs[j]);
+        if (res != nullptr) {
+          float rr[VEC];
+          load8(res + row * D + cid[j] * VEC, rr);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk)
+            xs[j][kk] = bf16_to_f32(f32_to_bf16(xs[j][kk] + rr[kk]));
+          store8(sum_out + row * D + cid[j] * VEC, xs[j]);
+        }
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) ssq += xs[j][kk] * xs[j][kk];
       }
@@ -122,6 +135,7 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                         const u16* __restrict__ w,
                         const float* __restrict__ rstd, u16* __restrict__ dx,
                         float* __restrict__ dw_part,  // [grid*G, D] fp32
+                        const u16* __restrict__ dadd,  // fused += residual grad
                         long long R, int D) {
   constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
@@ -175,6 +189,12 @@ void rmsnorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
           o[kk] = r * ds[j][kk] * wf[j][kk] - xs[j][kk] * coef;
           dwacc[j][kk] += ds[j][kk] * xs[j][kk] * r;
         }
+        if (dadd != nullptr) {
+          float da[VEC];
+          load8(dadd + row * D + cid[j] * VEC, da);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk) o[kk] += da[kk];
+        }
         store8(dxr + cid[j] * VEC, o);
       }
   }
@@ -210,6 +230,7 @@ void rmsnorm_bwd_wr_kernel(const u16* __restrict__ dy, const u16* __restrict__ x
                            const u16* __restrict__ w,
                            const float* __restrict__ rstd, u16* __restrict__ dx,
                            float* __restrict__ dw_part,  // [grid, D] fp32
+                           const u16* __restrict__ dadd,
                            long long R, int D) {
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];  // 16·D bytes
   const int nv = D / VEC;
@@ -259,6 +280,12 @@ void rmsnorm_bwd_wr_kernel(const u16* __restrict__ dy, const u16* __restrict__ x
           o[kk] = r * ds[j][kk] * wf[j][kk] - xs[j][kk] * coef;
           dwacc[j][kk] += ds[j][kk] * xs[j][kk] * r;
         }
+        if (dadd != nullptr) {
+          float da[VEC];
+          load8(dadd + row * D + cid[j] * VEC, da);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk) o[kk] += da[kk];
+        }
         store8(dxr + cid[j] * VEC, o);
       }
   }
@@ -288,6 +315,7 @@ void layernorm_bwd_wr_kernel(const u16* __restrict__ dy,
                              u16* __restrict__ dx,
                              float* __restrict__ dw_part,  // [grid, D]
                              float* __restrict__ db_part,  // [grid, D]
+                             const u16* __restrict__ dadd,
                              long long R, int D) {
   extern __shared__ __attribute__((aligned(16))) u16 w_lds[];  // 16·D bytes
   const int nv = D / VEC;
@@ -365,6 +393,12 @@ void layernorm_bwd_wr_kernel(const u16* __restrict__ dy,
           dwacc[j][kk] += ds[j][kk] * xh[j][kk];
           dbacc[j][kk] += ds[j][kk];
         }
+        if (dadd != nullptr) {
+          float da[VEC];
+          load8(dadd + row * D + cid[j] * VEC, da);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk) o[kk] += da[kk];
+        }
         store8(dxr + cid[j] * VEC, o);
       }
   }
@@ -402,6 +436,8 @@ void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
                           const u16* __restrict__ b, u16* __restrict__ y,
                           float* __restrict__ mean_out,
                           float* __restrict__ rstd_out,
+                          const u16* __restrict__ res,
+                          u16* __restrict__ sum_out,
                           long long R, int D, float eps) {
   constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
@@ -440,6 +476,14 @@ void layernorm_fwd_kernel(const u16* __restrict__ x, const u16* __restrict__ w,
     for (int j = 0; j < CH; ++j)
       if (live && act[j]) {
         load8(xr + cid[j] * VEC, xs[j]);
+        if (res != nullptr) {
+          float rr[VEC];
+          load8(res + row * D + cid[j] * VEC, rr);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk)
+            xs[j][kk] = bf16_to_f32(f32_to_bf16(xs[j][kk] + rr[kk]));
+          store8(sum_out + row * D + cid[j] * VEC, xs[j]);
+        }
 #pragma unroll
         for (int kk = 0; kk < VEC; ++kk) sum += xs[j][kk];
       }
@@ -483,6 +527,7 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
                           u16* __restrict__ dx,
                           float* __restrict__ dw_part,  // [grid*G, D]
                           float* __restrict__ db_part,  // [grid*G, D]
+                          const u16* __restrict__ dadd,
                           long long R, int D) {
   constexpr int TPG = BLOCK / G;
   __shared__ float lds[4];
@@ -542,6 +587,12 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
           o[kk] = r * (ds[j][kk] * wf[j][kk] - s1 - xh[j][kk] * s2);
           dwacc[j][kk] += ds[j][kk] * xh[j][kk];
           dbacc[j][kk] += ds[j][kk];
+        }
+        if (dadd != nullptr) {
+          float da[VEC];
+          load8(dadd + row * D + cid[j] * VEC, da);
+#pragma unroll
+          for (int kk = 0; kk < VEC; ++kk) o[kk] += da[kk];
         }
         store8(dxr + cid[j] * VEC, o);
       }
@@ -608,13 +659,14 @@ int acco_norm_bwd_grid(long long R, int D) {
 }
 
 void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
+                      const void* res, void* sum_out,
                       long long R, int D, float eps, hipStream_t s) {
   const int lds = D * sizeof(u16);
   const int G = groups_for(D);
   const int grid = fwd_blocks(R, G);
 #define L(CH, G) hipLaunchKernelGGL((rmsnorm_fwd_kernel<CH, G>), dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)x, (const u16*)w, (u16*)y, \
-    (float*)rstd, R, D, eps)
+    (float*)rstd, (const u16*)res, (u16*)sum_out, R, D, eps)
   if (G == 4) { L(1, 4); }
   else if (G == 2) { L(1, 2); }
   else switch (chunks_for(D)) {
@@ -626,7 +678,7 @@ void acco_rmsnorm_fwd(const void* x, const void* w, void* y, void* rstd,
 
 void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
                       const void* rstd, void* dx, void* dw_fp32,
-                      long long R, int D, hipStream_t s) {
+                      const void* dadd, long long R, int D, hipStream_t s) {
   if (use_wr(D)) {
     const int wlds = 16 * D;             // 4 fp32 partial rows (≥ weights)
     const int grid = wr_blocks(R);
@@ -634,12 +686,12 @@ void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
       hipLaunchKernelGGL((rmsnorm_bwd_wr_kernel<1>), dim3(grid), dim3(BLOCK),
                          wlds, s, (const u16*)dy, (const u16*)x,
                          (const u16*)w, (const float*)rstd, (u16*)dx,
-                         (float*)dw_fp32, R, D);
+                         (float*)dw_fp32, (const u16*)dadd, R, D);
     else
       hipLaunchKernelGGL((rmsnorm_bwd_wr_kernel<2>), dim3(grid), dim3(BLOCK),
                          wlds, s, (const u16*)dy, (const u16*)x,
                          (const u16*)w, (const float*)rstd, (u16*)dx,
-                         (float*)dw_fp32, R, D);
+                         (float*)dw_fp32, (const u16*)dadd, R, D);
     return;
   }
   const int lds = D * sizeof(u16);
@@ -647,7 +699,7 @@ void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
   const int grid = bwd_blocks(R, G);
 #define L(CH, G) hipLaunchKernelGGL((rmsnorm_bwd_kernel<CH, G>), dim3(grid), \
     dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, (const u16*)w, \
-    (const float*)rstd, (u16*)dx, (float*)dw_fp32, R, D)
+    (const float*)rstd, (u16*)dx, (float*)dw_fp32, (const u16*)dadd, R, D)
   if (G == 4) { L(1, 4); }
   else if (G == 2) { L(1, 2); }
   else switch (chunks_for(D)) {
@@ -658,14 +710,16 @@ void acco_rmsnorm_bwd(const void* dy, const void* x, const void* w,
 }
 
 void acco_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
-                        void* mean, void* rstd, long long R, int D, float eps,
+                        void* mean, void* rstd, const void* res,
+                        void* sum_out, long long R, int D, float eps,
                         hipStream_t s) {
   const int lds = 2 * D * sizeof(u16);
   const int G = groups_for(D);
   const int grid = fwd_blocks(R, G);
 #define L(CH, G) hipLaunchKernelGGL((layernorm_fwd_kernel<CH, G>), \
     dim3(grid), dim3(BLOCK), lds, s, (const u16*)x, (const u16*)w, \
-    (const u16*)b, (u16*)y, (float*)mean, (float*)rstd, R, D, eps)
+    (const u16*)b, (u16*)y, (float*)mean, (float*)rstd, (const u16*)res, \
+    (u16*)sum_out, R, D, eps)
   if (G == 4) { L(1, 4); }
   else if (G == 2) { L(1, 2); }
   else switch (chunks_for(D)) {
@@ -677,8 +731,8 @@ void acco_layernorm_fwd(const void* x, const void* w, const void* b, void* y,
 
 void acco_layernorm_bwd(const void* dy, const void* x, const void* w,
                         const void* mean, const void* rstd, void* dx,
-                        void* dw_fp32, void* db_fp32, long long R, int D,
-                        hipStream_t s) {
+                        void* dw_fp32, void* db_fp32, const void* dadd,
+                        long long R, int D, hipStream_t s) {
   if (use_wr(D)) {
     const int wlds = 16 * D;
     const int grid = wr_blocks(R);
@@ -687,13 +741,13 @@ void acco_layernorm_bwd(const void* dy, const void* x, const void* w,
                          dim3(BLOCK), wlds, s, (const u16*)dy, (const u16*)x,
                          (const u16*)w, (const float*)mean,
                          (const float*)rstd, (u16*)dx, (float*)dw_fp32,
-                         (float*)db_fp32, R, D);
+                         (float*)db_fp32, (const u16*)dadd, R, D);
     else
       hipLaunchKernelGGL((layernorm_bwd_wr_kernel<2>), dim3(grid),
                          dim3(BLOCK), wlds, s, (const u16*)dy, (const u16*)x,
                          (const u16*)w, (const float*)mean,
                          (const float*)rstd, (u16*)dx, (float*)dw_fp32,
-                         (float*)db_fp32, R, D);
+                         (float*)db_fp32, (const u16*)dadd, R, D);
     return;
   }
   const int lds = D * sizeof(u16);
@@ -702,7 +756,7 @@ void acco_layernorm_bwd(const void* dy, const void* x, const void* w,
 #define L(CH, G) hipLaunchKernelGGL((layernorm_bwd_kernel<CH, G>), \
     dim3(grid), dim3(BLOCK), lds, s, (const u16*)dy, (const u16*)x, \
     (const u16*)w, (const float*)mean, (const float*)rstd, (u16*)dx, \
-    (float*)dw_fp32, (float*)db_fp32, R, D)
+    (float*)dw_fp32, (float*)db_fp32, (const u16*)dadd, R, D)
   if (G == 4) { L(1, 4); }
   else if (G == 2) { L(1, 2); }
   else switch (chunks_for(D)) {

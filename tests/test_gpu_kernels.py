@@ -127,6 +127,49 @@ def test_ce_fwd_bwd():
     assert _close_bf16(logits.grad, l32.grad, atol=1e-3, rtol=5e-2)
 
 
+def test_add_norm_fused_fwd_bwd():
+    """Fused residual-add + norm (both families) vs the fp32 reference:
+    y, s values and dx/ds/dw(db) grads, including the dadd backward fold."""
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    torch.manual_seed(4)
+    for family in ("rms", "ln"):
+        D = 768
+        x = torch.randn(4, 33, D, device="cuda").bfloat16().requires_grad_(True)
+        res = torch.randn(4, 33, D, device="cuda").bfloat16().requires_grad_(True)
+        w = torch.randn(D, device="cuda").bfloat16().requires_grad_(True)
+        b = torch.randn(D, device="cuda").bfloat16().requires_grad_(True)
+        if family == "rms":
+            y, s = ops.add_rms_norm(x, res, w, 1e-5)
+        else:
+            y, s = ops.add_layer_norm(x, res, w, b, 1e-5)
+        # downstream uses BOTH outputs (residual chain + normed branch)
+        dout_y = torch.randn_like(y)
+        dout_s = torch.randn_like(s)
+        ((y.float() * dout_y.float()).sum()
+         + (s.float() * dout_s.float()).sum()).backward()
+
+        x32 = x.detach().float().requires_grad_(True)
+        r32 = res.detach().float().requires_grad_(True)
+        w32 = w.detach().float().requires_grad_(True)
+        b32 = b.detach().float().requires_grad_(True)
+        s32 = (x32 + r32)
+        if family == "rms":
+            y32 = torch_ref.rms_norm(s32, w32, 1e-5)
+        else:
+            y32 = torch_ref.layer_norm(s32, w32, b32, 1e-5)
+        ((y32 * dout_y.float()).sum() + (s32 * dout_s.float()).sum()).backward()
+
+        assert _close_bf16(y, y32.detach(), atol=3e-2, rtol=3e-2)
+        assert _close_bf16(s, s32.detach(), atol=1e-2, rtol=1e-2)
+        assert _close_bf16(x.grad, x32.grad, atol=3e-2, rtol=5e-2)
+        assert _close_bf16(res.grad, r32.grad, atol=3e-2, rtol=5e-2)
+        assert torch.allclose(w.grad.float(), w32.grad, atol=0.3, rtol=5e-2)
+        if family == "ln":
+            assert torch.allclose(b.grad.float(), b32.grad, atol=0.3,
+                                  rtol=5e-2)
+
+
 def test_ce_label_smoothed_fwd_bwd():
     """K9: label smoothing fused into the CE kernel vs the HF-parity
     torch reference (forward value + full dlogits)."""
