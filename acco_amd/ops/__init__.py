@@ -112,9 +112,13 @@ def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
 def add_rms_norm(x: torch.Tensor, res: Optional[torch.Tensor],
                  weight: torch.Tensor, eps: float):
     """(rmsnorm(x+res)·w, x+res) in ONE kernel pass (residual-add fusion);
-    res=None degenerates to a plain norm with s = x."""
+    res=None degenerates to a plain norm with s = x.
+    ACCO_NO_ADDNORM_FUSE=1 keeps the eager add + plain norm (same-box A/B)."""
     if res is None:
         return rms_norm(x, weight, eps), x
+    if os.environ.get("ACCO_NO_ADDNORM_FUSE") == "1":
+        s = x + res
+        return rms_norm(s, weight, eps), s
     if _use_ref(x, "add_rmsnorm_fwd"):
         s = x + res
         return torch_ref.rms_norm(s, weight, eps), s
@@ -127,6 +131,9 @@ def add_layer_norm(x: torch.Tensor, res: Optional[torch.Tensor],
     """(layernorm(x+res)·w+b, x+res) in ONE kernel pass; res=None → plain."""
     if res is None:
         return layer_norm(x, weight, bias, eps), x
+    if os.environ.get("ACCO_NO_ADDNORM_FUSE") == "1":
+        s = x + res
+        return layer_norm(s, weight, bias, eps), s
     if _use_ref(x, "add_layernorm_fwd"):
         s = x + res
         return torch_ref.layer_norm(s, weight, bias, eps), s

@@ -45,3 +45,36 @@ def test_bench_json_contract_ddp():
     d = _run(["--method", "ddp", "--steps", "2", "--warmup", "1"])
     assert d["config"]["method"] == "ddp"
     assert d["value"] > 0
+
+
+def test_bench_multirank_launch_contract():
+    """The driver's N>1 launch shape: torch.distributed.run --nnodes=1
+    --nproc-per-node 2 bench.py --gpus 2 ... (gloo on CPU here; RCCL on a
+    GPU node). Rank 0 must emit exactly one whole-job JSON line with
+    n_gpus=2 and the SUM-over-ranks token throughput."""
+    env = dict(os.environ)
+    env.pop("RANK", None); env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29611", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, cwd=REPO, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["global_batch"] == 2 * 8
+    assert d["value"] > 0
+
+
+def test_bench_refuses_mislabeled_gpus_flag():
+    """--gpus that contradicts WORLD_SIZE must refuse (exit 2), not emit a
+    mislabeled record."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--gpus", "4",
+         "--steps", "1", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 2
+    assert not [l for l in out.stdout.splitlines() if l.startswith("{")]
