@@ -25,6 +25,8 @@ extern "C" void acco_rmsnorm_bwd(const void*, const void*, const void*,
                                  const void*, void*, void*, const void*,
                                  long long, int, hipStream_t);
 extern "C" int acco_norm_bwd_grid(long long R, int D);
+extern "C" void acco_colsum(const void*, float*, long long, int, bool,
+                            hipStream_t);
 extern "C" void acco_layernorm_fwd(const void*, const void*, const void*,
                                    void*, void*, void*, const void*, void*,
                                    long long, int, float, hipStream_t);
@@ -101,6 +103,19 @@ void fused_adamw(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                           bf16, commit, (float)scale, (float)lr, (float)beta1,
                           (float)beta2, (float)eps, (float)weight_decay,
                           step + 1, stream.stream());
+}
+
+at::Tensor colsum(at::Tensor in) {
+  TORCH_CHECK(in.dim() == 2 && in.is_contiguous() && in.is_cuda());
+  TORCH_CHECK(in.scalar_type() == at::kBFloat16 ||
+              in.scalar_type() == at::kFloat);
+  const long long P = in.size(0);
+  const int D = (int)in.size(1);
+  auto out = at::zeros({D}, in.options().dtype(at::kFloat));
+  acco_colsum(in.data_ptr(), out.data_ptr<float>(), P, D,
+              in.scalar_type() == at::kBFloat16,
+              at::hip::getCurrentHIPStream().stream());
+  return out;
 }
 
 #define CHECK_BF16_CONTIG(t) \
@@ -222,7 +237,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
                    rstd.data_ptr(), dx.data_ptr(), dw_part.data_ptr(),
                    (dadd.defined() && dadd.numel() > 0) ? dadd.data_ptr() : nullptr, R, D,
                    cur_stream());
-  return {dx, dw_part.sum(0)};
+  return {dx, colsum(dw_part)};
 }
 
 // ---- LayerNorm
@@ -278,7 +293,7 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
                      dw_part.data_ptr(), db_part.data_ptr(),
                      (dadd.defined() && dadd.numel() > 0) ? dadd.data_ptr() : nullptr, R, D,
                      cur_stream());
-  return {dx, dw_part.sum(0), db_part.sum(0)};
+  return {dx, colsum(dw_part), colsum(db_part)};
 }
 
 // ---- RoPE ([B, S, H, D] contiguous)
@@ -488,6 +503,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_packed_bwd", &swiglu_packed_bwd);
   m.def("gelu_fwd", &gelu_fwd);
   m.def("gelu_bwd", &gelu_bwd);
+  m.def("colsum", &colsum);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd, py::arg("dy"), py::arg("x"),
         py::arg("w"), py::arg("rstd"), py::arg("dadd") = py::none());

@@ -609,6 +609,29 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
       }
 }
 
+// ------------------------------------------------- column sum (dim-0 sum)
+// out[d] = Σ_p in[p·D + d] for a [P, D] row-major matrix — the norm
+// dW/dB partial reduce and the bias-wgrad reduce. ATen's dim-0 reduce_kernel
+// runs this at ~150 GB/s (20.8 µs on [1024,768] fp32); coalesced row sweeps
+// with a P-split + fp32 atomics reach the roofline.
+template <typename Tin>
+__global__ __launch_bounds__(BLOCK)
+void colsum_kernel(const Tin* __restrict__ in, float* __restrict__ out,
+                   long long P, int D) {
+  const int d = blockIdx.x * BLOCK + threadIdx.x;    // one column per thread
+  if (d >= D) return;
+  const long long p0 = (long long)blockIdx.y * P / gridDim.y;
+  const long long p1 = (long long)(blockIdx.y + 1) * P / gridDim.y;
+  float acc = 0.0f;
+  for (long long p = p0; p < p1; ++p) {
+    if constexpr (sizeof(Tin) == 2)
+      acc += bf16_to_f32(((const u16*)in)[p * D + d]);
+    else
+      acc += ((const float*)in)[p * D + d];
+  }
+  atomicAdd(out + d, acc);
+}
+
 int groups_for(int D) {
   const int nv = D / VEC;
   if (nv <= 64) return 4;      // one wave per row
@@ -650,6 +673,23 @@ int wr_blocks(long long R) {
 }  // namespace
 
 extern "C" {
+
+void acco_colsum(const void* in, float* out, long long P, int D,
+                 bool in_is_bf16, hipStream_t s) {
+  const int xblocks = (D + BLOCK - 1) / BLOCK;
+  // split P so the grid covers the chip even at small D (out is zeroed by
+  // the caller; partial sums combine via fp32 atomics)
+  int ysplit = (int)(256 / (xblocks < 1 ? 1 : xblocks));
+  if (ysplit < 1) ysplit = 1;
+  if ((long long)ysplit > P) ysplit = (int)(P < 1 ? 1 : P);
+  dim3 grid(xblocks, ysplit);
+  if (in_is_bf16)
+    hipLaunchKernelGGL((colsum_kernel<u16>), grid, dim3(BLOCK), 0, s,
+                       (const u16*)in, out, P, D);
+  else
+    hipLaunchKernelGGL((colsum_kernel<float>), grid, dim3(BLOCK), 0, s,
+                       (const float*)in, out, P, D);
+}
 
 // number of fp32 partial rows the bwd kernels emit (scratch allocation)
 int acco_norm_bwd_grid(long long R, int D) {

@@ -127,6 +127,23 @@ def test_ce_fwd_bwd():
     assert _close_bf16(logits.grad, l32.grad, atol=1e-3, rtol=5e-2)
 
 
+def test_colsum():
+    """colsum (dim-0 sum of [P, D]) vs torch, bf16 and fp32 inputs."""
+    from acco_amd import ops
+    ext = ops.hip_ext()
+    torch.manual_seed(2)
+    for dt, atol in ((torch.float32, 1e-3), (torch.bfloat16, 0.5)):
+        x = torch.randn(1024, 768, device="cuda", dtype=dt)
+        got = ext.colsum(x)
+        ref = x.float().sum(0)
+        assert got.dtype == torch.float32
+        assert torch.allclose(got, ref, atol=atol, rtol=1e-3), \
+            (got - ref).abs().max()
+    # odd shapes / D not a multiple of the block
+    x = torch.randn(37, 100, device="cuda")
+    assert torch.allclose(ext.colsum(x), x.sum(0), atol=1e-3, rtol=1e-3)
+
+
 def test_add_norm_fused_fwd_bwd():
     """Fused residual-add + norm (both families) vs the fp32 reference:
     y, s values and dx/ds/dw(db) grads, including the dadd backward fold."""
