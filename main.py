@@ -11,6 +11,7 @@ data=openwebtext / data=alpaca on a machine with HF hub access).
 """
 
 import logging
+import os
 import sys
 
 import torch
@@ -47,8 +48,87 @@ def build_datasets(cfg, model):
     return ds["train"], ds["test"], tokenizer
 
 
+def _resolve_run_dir(template: str) -> str:
+    """Hydra-style date templating (reference config/config.yaml:10-12):
+    ${now:%Y-%m-%d} etc. expand against the launch time."""
+    import datetime
+    import re
+    now = datetime.datetime.now()
+    return re.sub(r"\$\{now:([^}]*)\}", lambda m: now.strftime(m.group(1)),
+                  template)
+
+
+def _absolutize_local_paths(cfg) -> None:
+    """Run dirs chdir (Hydra semantics); local corpus/tokenizer paths in the
+    data group must survive that."""
+    for key in ("path", "tokenizer"):
+        v = cfg.data.get(key)
+        if isinstance(v, str) and os.path.exists(v):
+            cfg.data[key] = os.path.abspath(v)
+
+
+def _enter_run_dir(cfg, job_subdir=None):
+    """Create + chdir into the templated run dir (the reference's Hydra
+    run-management: every run writes results.csv / checkpoints / scalars
+    into its own date-stamped directory). run_dir=. disables."""
+    template = cfg.get("run_dir") or "."
+    if template in (".", "none", ""):
+        return None
+    d = _resolve_run_dir(template)
+    if job_subdir is not None:
+        d = os.path.join(d, str(job_subdir))
+    os.makedirs(d, exist_ok=True)
+    _absolutize_local_paths(cfg)
+    prev = os.getcwd()
+    os.chdir(d)
+    logger.info("run dir: %s", os.getcwd())
+    return prev
+
+
+def _expand_multirun(overrides):
+    """Hydra -m sweep: `a=1,2 b=x,y` → the cross product of run configs."""
+    import itertools
+    fixed, sweeps = [], []
+    for ov in overrides:
+        key, _, val = ov.partition("=")
+        if "," in val and not val.startswith(("[", "{")):
+            sweeps.append([(key, v) for v in val.split(",")])
+        else:
+            fixed.append(ov)
+    if not sweeps:
+        return [list(overrides)]
+    return [fixed + [f"{k}={v}" for k, v in combo]
+            for combo in itertools.product(*sweeps)]
+
+
+def run_one(overrides, job_subdir=None):
+    cfg = load_config(overrides)
+    prev = _enter_run_dir(cfg, job_subdir)
+    try:
+        _train(cfg)
+    finally:
+        if prev is not None:
+            os.chdir(prev)
+
+
 def main(argv=None):
-    cfg = load_config(argv if argv is not None else sys.argv[1:])
+    argv = list(argv if argv is not None else sys.argv[1:])
+    multirun = False
+    for flag in ("-m", "--multirun"):
+        if flag in argv:
+            argv.remove(flag)
+            multirun = True
+    if multirun:
+        jobs = _expand_multirun(argv)
+        logger.info("multirun: %d jobs", len(jobs))
+        for i, job in enumerate(jobs):
+            logger.info("=== multirun job %d: %s", i, job)
+            run_one(job, job_subdir=i)
+        return
+    run_one(argv)
+
+
+def _train(cfg):
     torch.manual_seed(42)
     if torch.cuda.is_available():
         torch.cuda.manual_seed_all(42)  # reference main.py:28

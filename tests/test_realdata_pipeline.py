@@ -71,6 +71,7 @@ def test_train_and_perplexity_on_local_corpus(small_corpus, tmp_path,
     corpus_args = [f"data.path={small_corpus}",
                    f"data.tokenizer={TOKENIZER}"]
     train_main.main(["train=dpu", "data=localtext", "model=gptneo",
+                     "run_dir=.",
                      "train.nb_steps_tot=4", "train.n_warmup_steps=0",
                      "train.eval=false", "train.save=true",
                      "train.use_mixed_precision=false",
