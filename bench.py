@@ -94,8 +94,9 @@ def main():
 
     cuda = torch.cuda.is_available()
     if cuda:
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        dev_idx = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
         backend = "nccl"
     else:
         device = torch.device("cpu")
