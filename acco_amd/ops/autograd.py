@@ -234,7 +234,12 @@ class AttnQKVPackedFn(torch.autograd.Function):
         dkq, dvq = ext.attn_bwd_packed(roped, dO, lse, delta, dqkv, H, Hkv,
                                        D, scale, window, q_off, k_off, v_off)
         rep = H // Hkv
-        if rep > 1:
+        if hasattr(ext, "attn_gqa_reduce"):
+            # one pass: group-sum over the rep query heads + bf16 cast +
+            # strided write into the packed grad's k|v sections (replaces
+            # two dim-3 sums, two casts and two strided copies per layer)
+            ext.attn_gqa_reduce(dkq, dvq, dqkv, Hkv, rep, D, k_off, v_off)
+        elif rep > 1:
             dk = dkq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
             dv = dvq.view(B, S, Hkv, rep, D).sum(3, dtype=torch.float32)
             dqkv[..., k_off:k_off + Hkv * D] = dk.reshape(B, S, Hkv * D).bfloat16()

@@ -66,6 +66,9 @@ extern "C" void acco_attn_bwd_dq(const void*, const void*, const void*,
                                  hipStream_t);
 extern "C" void acco_attn_delta(const void*, const void*, float*, long long,
                                 int, int, int, hipStream_t);
+extern "C" void acco_attn_gqa_reduce(const void*, const void*, void*,
+                                     long long, int, int, int, long long,
+                                     long long, long long, hipStream_t);
 extern "C" void acco_attn_bwd_dkv(const void*, const void*, const void*,
                                   const void*, const float*, const float*,
                                   void*, void*, int, int, int, int, int,
@@ -418,6 +421,19 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// GQA group-reduce of per-query-head dK/dV straight into the packed grad
+void attn_gqa_reduce(at::Tensor dkq, at::Tensor dvq, at::Tensor dqkv,
+                     int64_t Hkv, int64_t rep, int64_t D, int64_t k_off,
+                     int64_t v_off) {
+  CHECK_BF16_CONTIG(dkq); CHECK_BF16_CONTIG(dvq); CHECK_BF16_CONTIG(dqkv);
+  TORCH_CHECK(D % 8 == 0);
+  const long long T = dkq.size(0) * dkq.size(1);
+  const long long W = dqkv.size(2);
+  acco_attn_gqa_reduce(dkq.data_ptr(), dvq.data_ptr(), dqkv.data_ptr(), T,
+                       (int)Hkv, (int)rep, (int)D, W, k_off, v_off,
+                       cur_stream());
+}
+
 // ---- experimental NT GEMM (C = A · B^T, bf16; bench/refcheck only)
 at::Tensor gemm_nt(at::Tensor A, at::Tensor B) {
   CHECK_BF16_CONTIG(A); CHECK_BF16_CONTIG(B);
@@ -525,6 +541,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_delta", &attn_delta);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_gqa_reduce", &attn_gqa_reduce);
   m.def("gemm_nt", &gemm_nt);
   m.def("rope_packed", &rope_packed);
   m.def("attn_fwd_packed", &attn_fwd_packed);
