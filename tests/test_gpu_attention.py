@@ -187,3 +187,45 @@ def test_packed_qkv_core_llama_d128_matches_standard():
     cos = torch.nn.functional.cosine_similarity(g1[:n].float(),
                                                 g2[:n].float(), dim=0)
     assert cos > 0.995, f"grad cosine {cos}"
+
+
+def test_attn_defer_max_rescale_forced():
+    """T13 hazard test (guide §5.4 rule 26): the defer-max branch is rare on
+    random data — force a LATE rescale by spiking K rows deep into the
+    sequence so the running max jumps past the threshold at chosen tiles,
+    after O has accumulated. Checked against the fp32 reference."""
+    from acco_amd import ops
+    from acco_amd.ops import torch_ref
+    torch.manual_seed(7)
+    B, S, H, Hkv, D = 2, 1024, 8, 4, 64
+    q = (torch.randn(B, S, H, D, device="cuda") * 0.5).bfloat16()
+    k = (torch.randn(B, S, Hkv, D, device="cuda") * 0.5).bfloat16()
+    v = (torch.randn(B, S, Hkv, D, device="cuda") * 0.5).bfloat16()
+    # spike several K rows at different tiles: q·k for these rows ≫ any
+    # other score (|q·k| ≲ scale·D·0.25 ≈ 2 normally; the spike gives
+    # ≈ 0.125·8·64·0.5·~q ≫ 8/ln2 growth), so every q row ≥ the spike
+    # position must rescale its accumulated O there
+    for pos in (300, 650, 900):
+        k[:, pos, :, :] = 8.0
+    q.requires_grad_(True)
+    k.requires_grad_(True)
+    v.requires_grad_(True)
+
+    out = ops.causal_attention(q, k, v)
+    ref = torch_ref.causal_attention(q.detach().float(), k.detach().float(),
+                                     v.detach().float())
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2), \
+        (out.float() - ref).abs().max()
+
+    dO = torch.randn_like(out)
+    out.backward(dO)
+    q32 = q.detach().float().requires_grad_(True)
+    k32 = k.detach().float().requires_grad_(True)
+    v32 = v.detach().float().requires_grad_(True)
+    torch_ref.causal_attention(q32, k32, v32).backward(dO.float())
+    for name, got, want in [("dq", q.grad, q32.grad),
+                            ("dk", k.grad, k32.grad),
+                            ("dv", v.grad, v32.grad)]:
+        e = (got.float() - want).abs().max().item()
+        assert torch.allclose(got.float(), want, atol=8e-2, rtol=8e-2), \
+            f"{name} err {e}"
