@@ -61,6 +61,19 @@ def attach_grad_arena(model: nn.Module, dtype: torch.dtype, device,
     return arena
 
 
+def repoint_params(model: nn.Module, arena: torch.Tensor) -> None:
+    """Re-home every parameter's data view onto `arena` (same flat order as
+    flatten_params) WITHOUT copying — the arena-swap optimization: after an
+    ACCO com round the com buffer already holds the new parameters, so the
+    params-arena and com-buffer swap ROLES instead of moving 2·N bytes."""
+    off = 0
+    with torch.no_grad():
+        for p in param_order(model):
+            num = p.numel()
+            p.data = arena[off:off + num].view_as(p)
+            off += num
+
+
 def live_numel(model: nn.Module) -> int:
     return sum(p.numel() for p in model.parameters())
 

@@ -66,15 +66,15 @@ class DecoupledTrainer:
         self.spec = ShardSpec.build(n_live, self.world_size,
                                     buckets=int(getattr(args, "comm_buckets", 8) or 8))
         self.model = model
-        self.params = arena.flatten_params(model, self.dtype, self.device,
-                                           pad_to=self.spec.total)
+        self._params = arena.flatten_params(model, self.dtype, self.device,
+                                            pad_to=self.spec.total)
         self.grads = arena.attach_grad_arena(model, self.dtype, self.device,
                                              pad_to=self.spec.total)
         self.n_live = n_live
 
         self.comm = CommBackend(self.device)
         # C1: average random-init weights across ranks (reference :180)
-        self.comm.all_reduce_avg(self.params)
+        self.comm.all_reduce_avg(self._params)
 
         # fused QKV / gate-up GEMMs over arena-adjacent weight views — in
         # EVERY mode, including DDP: the native DDP's bucket readiness is
@@ -82,7 +82,7 @@ class DecoupledTrainer:
         # feeds its backward-overlap launches too. ACCO vs DDP therefore
         # compares the algorithms on identical compute kernels.
         from acco_amd.models.fuse import install_fused_projections
-        install_fused_projections(model, self.params, self.grads)
+        install_fused_projections(model, self._params, self.grads)
 
         # per-layer activation recompute (8B-on-288GB batch headroom)
         if getattr(args, "activation_checkpointing", False):
@@ -117,14 +117,14 @@ class DecoupledTrainer:
 
         if args.method_name == "ddp" or getattr(args, "run_baseline_ddp", False):
             self.loss_div = float(args.n_grad_accumulation)
-            self.opt.init_master_from_buffer(self.params)
-            self.ddp = NativeZeroDDP(model, self.params, self.grads,
+            self.opt.init_master_from_buffer(self._params)
+            self.ddp = NativeZeroDDP(model, self._params, self.grads,
                                      self.n_live, self.spec, self.comm,
                                      self.rank, self.opt)
             self.engine = None
         else:
             self.engine = AccoEngine(
-                params_arena=self.params, grads_arena=self.grads,
+                params_arena=self._params, grads_arena=self.grads,
                 n_live=self.n_live, spec=self.spec, comm=self.comm,
                 rank=self.rank, device=self.device, opt=self.opt,
                 sched=self.sched,
@@ -134,14 +134,24 @@ class DecoupledTrainer:
                 grad_reduce_dtype=getattr(args, "grad_reduce_dtype", None),
                 log=self.log)
             # seed fp32 master from current (averaged) params
-            self.opt.init_master_from_buffer(self.params)
+            self.opt.init_master_from_buffer(self._params)
             self.engine.on_round_complete = self._on_round_complete
+            # zero-copy params⇄buffer handover (engine/acco.py)
+            self.engine.enable_arena_swap(model, self.grads)
             self.ddp = None
 
         self.t_beg = time.time()
         self.t_last_epoch = self.t_beg
         self._time_checkpoint = time.time()
         self._last_eval = 0
+
+    @property
+    def params(self):
+        """The LIVE flat parameter arena: under the engine's arena-swap
+        the params/com-buffer tensors trade roles every round, so read
+        through the engine rather than the construction-time reference."""
+        eng = getattr(self, "engine", None)
+        return eng.params if eng is not None else self._params
 
     # ------------------------------------------------------------- data
 

@@ -226,6 +226,28 @@ def install_fused_projections(model: nn.Module, params_arena: torch.Tensor,
     return count
 
 
+_FUSE_ATTRS = ("_fused_qkv", "_fused_gate_up", "_fused_kvq", "_arena_fuse")
+
+
+def snapshot_fused_views(model: nn.Module):
+    """Capture the installed arena views (one set per arena for the
+    arena-swap optimization: install against each arena, snapshot, then
+    flip between snapshots with apply_fused_views — no per-round
+    re-derivation)."""
+    snap = []
+    for mod in model.modules():
+        for attr in _FUSE_ATTRS:
+            v = getattr(mod, attr, None)
+            if v is not None:
+                snap.append((mod, attr, v))
+    return snap
+
+
+def apply_fused_views(snapshot) -> None:
+    for mod, attr, v in snapshot:
+        setattr(mod, attr, v)
+
+
 def notify_producers(model: nn.Module):
     """Enumerate the grad-arena ranges filled by the in-place-dW notify path
     (fused groups + wrapped singles) as ``[(offset, numel)]``, plus the set

@@ -194,6 +194,7 @@ def main():
                 state["t1"] = time.time(); state["mb1"] = eng.micro_steps
 
         eng.on_round_boundary = boundary
+        eng.enable_arena_swap(model, grads)
         if args.method == "acco":
             eng.train_acco(nb_grad_tot=1 << 60, n_warmup_steps=0,
                            max_rounds=warmup + steps)
