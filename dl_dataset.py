@@ -34,8 +34,16 @@ def main(argv=None):
             fn, batched=True, remove_columns=ds[split].column_names,
             num_proc=cfg.train.dataloader_num_workers or 1)
     packed = datasets.DatasetDict(out)
-    packed.save_to_disk(out_dir)
-    print(f"saved packed dataset to {out_dir}")
+    # out_shards=N splits each split into N arrow shards (parallel loaders /
+    # per-rank file mapping on big corpora); 0/absent keeps one file
+    shards = int(cfg.get("out_shards", 0) or 0)
+    if shards > 1:
+        packed.save_to_disk(out_dir,
+                            num_shards={s: shards for s in packed})
+    else:
+        packed.save_to_disk(out_dir)
+    print(f"saved packed dataset to {out_dir}"
+          + (f" ({shards} shards/split)" if shards > 1 else ""))
 
 
 if __name__ == "__main__":

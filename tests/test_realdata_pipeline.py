@@ -53,13 +53,16 @@ def test_dl_dataset_offline_pack(small_corpus, tmp_path, monkeypatch):
     dl_dataset.main(["data=localtext", "model=gptneo",
                      f"data.path={small_corpus}",
                      "train.max_length=128",
+                     "out_shards=2",
                      f"out_dir={out}"])
+    import glob as _glob
     import datasets
     packed = datasets.load_from_disk(out)
     assert "train" in packed and "test" in packed
     row = packed["train"][0]["input_ids"]
     assert len(row) == 128
     assert all(0 <= t < 8192 for t in row)
+    assert len(_glob.glob(out + "/train/data-*.arrow")) == 2
 
 
 def test_train_and_perplexity_on_local_corpus(small_corpus, tmp_path,
