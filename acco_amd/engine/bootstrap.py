@@ -68,6 +68,10 @@ def detect_topology() -> dict:
 
 def init_distributed(backend: str | None = None,
                      timeout_s: int = 600) -> DistContext:
+    # the host driver on this pool only supports dmabuf IPC; without this
+    # RCCL cross-process CUDA-tensor sharing fails with
+    # `hipIpcGetMemHandle: invalid argument`
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     topo = detect_topology()
     cuda = torch.cuda.is_available()
     if backend is None:

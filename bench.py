@@ -80,6 +80,8 @@ def make_batch_pool(n_pool, batch, seq, vocab, device, seed):
 
 def main():
     args = parse_args()
+    # dmabuf IPC (see engine/bootstrap.py) — required for multi-rank RCCL
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     env_world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))

@@ -47,6 +47,12 @@ def test_bench_json_contract_ddp():
     assert d["value"] > 0
 
 
+def test_bench_json_contract_dpu():
+    d = _run(["--method", "dpu", "--steps", "2", "--warmup", "1"])
+    assert d["config"]["method"] == "dpu"
+    assert d["value"] > 0
+
+
 def test_bench_multirank_launch_contract():
     """The driver's N>1 launch shape: torch.distributed.run --nnodes=1
     --nproc-per-node 2 bench.py --gpus 2 ... (gloo on CPU here; RCCL on a
