@@ -84,11 +84,14 @@ def _worker_acco(rank, world, port, tmpdir, n_warmup):
                      sched=sched, forward_backward=forward_backward,
                      next_batch=next_batch, n_grad_accumulation=N_ACC)
     opt.init_master_from_buffer(params)
+    # arena-swap ON: the oracle replay then verifies the zero-copy
+    # handover's exact algebra (trace-driven, so thread-timing-independent)
+    eng.enable_arena_swap(model, grads)
     eng.trace = []
     eng.train_acco(TARGET, n_warmup_steps=n_warmup)
 
     torch.save({
-        "params": params[:D].clone(),
+        "params": eng.params[:D].clone(),
         "trace": eng.trace,
         "round_idx": eng.round_idx,
         "count_grad_tot": eng.count_grad_tot,
