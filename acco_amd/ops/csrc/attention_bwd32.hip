@@ -31,7 +31,15 @@ constexpr int BT = QW * NW;       // 256 rows per workgroup
   __builtin_amdgcn_mfma_f32_32x32x16_bf16((a), (b), (c), 0, 0, 0)
 
 ACCO_DEV unsigned pack_bf16_(float lo, float hi) {
-  return (unsigned)f32_to_bf16(lo) | ((unsigned)f32_to_bf16(hi) << 16);
+  // v_cvt_pk_bf16_f32 (RNE, no builtin on gfx950 — guide T12): one
+  // instruction replaces ~9 VALU of manual round-to-nearest-even
+  // bit-twiddling per packed dword. The trailing s_nop 1 covers the
+  // VALU-write → v_permlane32_swap hazard window for the consumer
+  // (guide T21 hazard note; hipcc pads nothing inside asm).
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+      : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
 }
 
 // C-layout floats (16 regs of one 32-row sub-tile) → A-fragment for K-step

@@ -16,13 +16,13 @@ ACCO_DEV float bf16_to_f32(unsigned short u) {
 }
 
 ACCO_DEV unsigned short f32_to_bf16(float f) {
-  union { float f; unsigned int i; } c;
-  c.f = f;
-  // round-to-nearest-even, matching PyTorch's float->bfloat16 cast
-  unsigned int x = c.i;
-  unsigned int rounding_bias = 0x7FFF + ((x >> 16) & 1);
-  x += rounding_bias;
-  return (unsigned short)(x >> 16);
+  // __float2bfloat16 is round-to-nearest-even on ROCm (identical to
+  // PyTorch's cast and to the manual +0x7FFF/odd-bit twiddle this
+  // replaced); the compiler pairs adjacent casts into v_cvt_pk_bf16_f32 —
+  // ~4 VALU ops fewer per element in pack-heavy epilogues
+  union { __hip_bfloat16 b; unsigned short u; } c;
+  c.b = __float2bfloat16(f);
+  return c.u;
 }
 
 // Grid sizing for memory-bound elementwise kernels (guide Guideline 11):
