@@ -614,22 +614,51 @@ void layernorm_bwd_kernel(const u16* __restrict__ dy, const u16* __restrict__ x,
 // dW/dB partial reduce and the bias-wgrad reduce. ATen's dim-0 reduce_kernel
 // runs this at ~150 GB/s (20.8 µs on [1024,768] fp32); coalesced row sweeps
 // with a P-split + fp32 atomics reach the roofline.
+// VECW consecutive columns per thread, 16-byte row loads (guide G13: a
+// scalar-load version of this ran at 0.5 TB/s and regressed the GPT-Neo
+// bias-wgrad path 4x vs ATen before being caught in the step profile)
 template <typename Tin>
 __global__ __launch_bounds__(BLOCK)
 void colsum_kernel(const Tin* __restrict__ in, float* __restrict__ out,
                    long long P, int D) {
-  const int d = blockIdx.x * BLOCK + threadIdx.x;    // one column per thread
-  if (d >= D) return;
+  constexpr int VECW = (sizeof(Tin) == 2) ? 8 : 4;   // 16 B per lane
+  const int d0 = (blockIdx.x * BLOCK + threadIdx.x) * VECW;
+  if (d0 >= D) return;
   const long long p0 = (long long)blockIdx.y * P / gridDim.y;
   const long long p1 = (long long)(blockIdx.y + 1) * P / gridDim.y;
-  float acc = 0.0f;
-  for (long long p = p0; p < p1; ++p) {
-    if constexpr (sizeof(Tin) == 2)
-      acc += bf16_to_f32(((const u16*)in)[p * D + d]);
-    else
-      acc += ((const float*)in)[p * D + d];
+  float acc[VECW];
+#pragma unroll
+  for (int k = 0; k < VECW; ++k) acc[k] = 0.0f;
+  if (d0 + VECW <= D) {
+    for (long long p = p0; p < p1; ++p) {
+      if constexpr (sizeof(Tin) == 2) {
+        const u16* row = (const u16*)in + p * D + d0;
+        ushort4 a = reinterpret_cast<const ushort4*>(row)[0];
+        ushort4 b = reinterpret_cast<const ushort4*>(row)[1];
+        acc[0] += bf16_to_f32(a.x); acc[1] += bf16_to_f32(a.y);
+        acc[2] += bf16_to_f32(a.z); acc[3] += bf16_to_f32(a.w);
+        acc[4] += bf16_to_f32(b.x); acc[5] += bf16_to_f32(b.y);
+        acc[6] += bf16_to_f32(b.z); acc[7] += bf16_to_f32(b.w);
+      } else {
+        const float4 v =
+            *reinterpret_cast<const float4*>((const float*)in + p * D + d0);
+        acc[0] += v.x; acc[1] += v.y; acc[2] += v.z; acc[3] += v.w;
+      }
+    }
+  } else {
+    for (long long p = p0; p < p1; ++p)      // ragged tail columns
+#pragma unroll
+      for (int k = 0; k < VECW; ++k)
+        if (d0 + k < D) {
+          if constexpr (sizeof(Tin) == 2)
+            acc[k] += bf16_to_f32(((const u16*)in)[p * D + d0 + k]);
+          else
+            acc[k] += ((const float*)in)[p * D + d0 + k];
+        }
   }
-  atomicAdd(out + d, acc);
+#pragma unroll
+  for (int k = 0; k < VECW; ++k)
+    if (d0 + k < D) atomicAdd(out + d0 + k, acc[k]);
 }
 
 int groups_for(int D) {
@@ -676,10 +705,11 @@ extern "C" {
 
 void acco_colsum(const void* in, float* out, long long P, int D,
                  bool in_is_bf16, hipStream_t s) {
-  const int xblocks = (D + BLOCK - 1) / BLOCK;
+  const int vecw = in_is_bf16 ? 8 : 4;
+  const int xblocks = (D + BLOCK * vecw - 1) / (BLOCK * vecw);
   // split P so the grid covers the chip even at small D (out is zeroed by
   // the caller; partial sums combine via fp32 atomics)
-  int ysplit = (int)(256 / (xblocks < 1 ? 1 : xblocks));
+  int ysplit = (int)(512 / (xblocks < 1 ? 1 : xblocks));
   if (ysplit < 1) ysplit = 1;
   if ((long long)ysplit > P) ysplit = (int)(P < 1 ? 1 : P);
   dim3 grid(xblocks, ysplit);
