@@ -60,6 +60,11 @@ class ScalarLogger:
         if self._tb is not None:
             self._tb.add_scalar(tag, value, step)
 
+    def add_histogram(self, tag: str, values, step: int) -> None:
+        """Distribution logging (grad/param histograms) — TB-only channel."""
+        if self._tb is not None:
+            self._tb.add_histogram(tag, values, step)
+
     def log_training(self, opt_step: int, n_grads: int, rank: int, loss: float,
                      eval_loss: Optional[float], t_beg: float) -> None:
         """Scalar-name schema mirrors reference log_to_tensorboard

@@ -95,5 +95,38 @@ class EventFileWriter:
         self._write_record(_encode_event(time.time(), step, tag,
                                          float(value)))
 
+    def add_histogram(self, tag: str, values, step: int,
+                      bins: int = 30) -> None:
+        """Summary.Value.histo (HistogramProto) — param/grad distribution
+        logging the reference gets from SummaryWriter; linear bins."""
+        vals = [float(v) for v in values]
+        if not vals:
+            return
+        lo, hi = min(vals), max(vals)
+        if hi == lo:
+            hi = lo + 1e-12
+        width = (hi - lo) / bins
+        counts = [0.0] * bins
+        for v in vals:
+            idx = int((v - lo) / width)
+            counts[min(idx, bins - 1)] += 1.0
+        limits = [lo + width * (i + 1) for i in range(bins)]
+        histo = (_field(1, 1) + struct.pack("<d", lo)
+                 + _field(2, 1) + struct.pack("<d", hi)
+                 + _field(3, 1) + struct.pack("<d", float(len(vals)))
+                 + _field(4, 1) + struct.pack("<d", sum(vals))
+                 + _field(5, 1) + struct.pack("<d", sum(v * v for v in vals)))
+        for num, seq in ((6, limits), (7, counts)):     # packed doubles
+            body = b"".join(struct.pack("<d", x) for x in seq)
+            histo += _field(num, 2) + _varint(len(body)) + body
+        t = tag.encode()
+        val = (_field(1, 2) + _varint(len(t)) + t
+               + _field(7, 2) + _varint(len(histo)) + histo)  # Value.histo
+        summ = _field(1, 2) + _varint(len(val)) + val
+        ev = (_field(1, 1) + struct.pack("<d", time.time())
+              + _field(2, 0) + _varint(step & 0xFFFFFFFFFFFFFFFF)
+              + _field(5, 2) + _varint(len(summ)) + summ)
+        self._write_record(ev)
+
     def close(self) -> None:
         self._f.close()
