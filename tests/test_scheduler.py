@@ -44,3 +44,31 @@ def test_clamps_past_total():
     s = LRSchedule(1.0, 0, 10, "cosine")
     assert s.factor(20) == pytest.approx(0.0, abs=1e-12)
     assert not math.isnan(s.factor(20))
+
+
+# --------- property: advance() is additive for any chunking of the steps
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from acco_amd.engine.scheduler import LRSchedule
+
+
+@settings(max_examples=150, deadline=None)
+@given(chunks=st.lists(st.integers(1, 50), min_size=1, max_size=12),
+       warmup=st.integers(0, 40),
+       total=st.integers(1, 400),
+       kind=st.sampled_from(["cosine", "linear", "constant"]))
+def test_advance_additivity(chunks, warmup, total, kind):
+    """Heterogeneous com rounds advance by varying global grad counts; the
+    LR must depend only on the cumulative count, never the chunking."""
+    a = LRSchedule(3e-4, warmup, total, kind)
+    for c in chunks:
+        a.advance(c)
+    b = LRSchedule(3e-4, warmup, total, kind)
+    b.advance(sum(chunks))
+    assert a.lr() == b.lr()
+    sd = a.state_dict()
+    c2 = LRSchedule(3e-4, warmup, total, kind)
+    c2.load_state_dict(sd)
+    assert c2.lr() == a.lr()
