@@ -80,3 +80,12 @@ def test_gloo_bucket_collectives_ws2():
     tmpdir = run_distributed(_worker_collectives, 2)
     for r in range(2):
         assert os.path.exists(os.path.join(tmpdir, f"ok_{r}.pt"))
+
+
+def test_gloo_bucket_collectives_ws3_ragged():
+    """world=3 with n=60 does not divide evenly (seg rounds 60/9 up to 8,
+    12 padding elems) — exercises the ragged/padded shard geometry through
+    the real collectives, not just ShardSpec math."""
+    tmpdir = run_distributed(_worker_collectives, 3)
+    for r in range(3):
+        assert os.path.exists(os.path.join(tmpdir, f"ok_{r}.pt"))
