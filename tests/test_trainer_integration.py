@@ -62,8 +62,9 @@ def _worker_dpu(rank, world, port, tmpdir):
 def _check(tmpdir, method, world=2):
     res = [torch.load(os.path.join(tmpdir, f"p_{method}_{r}.pt"),
                       weights_only=False) for r in range(world)]
-    assert torch.equal(res[0]["params"], res[1]["params"]), \
-        f"{method}: ranks diverged"
+    for r in range(1, world):
+        assert torch.equal(res[0]["params"], res[r]["params"]), \
+            f"{method}: rank {r} diverged"
     assert torch.isfinite(res[0]["params"]).all()
     # results.csv written by rank 0
     assert os.path.exists(os.path.join(tmpdir, "results.csv"))
@@ -77,6 +78,14 @@ def test_trainer_ddp_ws2():
 def test_trainer_acco_ws2():
     tmpdir = run_distributed(_worker_acco, 2, timeout=300)
     _check(tmpdir, "acco")
+
+
+def test_trainer_acco_ws3_ragged():
+    """world=3: the flat-arena shards do not divide evenly — the full
+    trainer (engine + collectives + sharded AdamW) must still keep every
+    rank's params bitwise identical."""
+    tmpdir = run_distributed(_worker_acco, 3, timeout=300)
+    _check(tmpdir, "acco", world=3)
 
 
 def test_trainer_dpu_ws2():
