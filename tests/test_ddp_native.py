@@ -147,3 +147,13 @@ def test_native_ddp_matches_adamw_oracle_ws2():
     flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
     assert torch.allclose(res[0], flat, atol=1e-6, rtol=1e-5), \
         (res[0] - flat).abs().max()
+
+
+def test_native_ddp_fused_ws3_ragged():
+    """world=3 (shards don't divide evenly): coverage-based bucket launch +
+    ragged sharded AdamW + all-gather keep all three ranks bitwise equal."""
+    d = run_distributed(_worker_ddp_fused, 3, args=(True,), timeout=240)
+    ps = [torch.load(os.path.join(d, f"p_f_{r}.pt"), weights_only=False)
+          for r in range(3)]
+    assert torch.equal(ps[0], ps[1]) and torch.equal(ps[0], ps[2])
+    assert torch.isfinite(ps[0]).all()
