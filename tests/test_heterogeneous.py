@@ -93,3 +93,22 @@ def test_acco_handoff_checksum_ws2():
     res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
                       weights_only=False) for r in range(2)]
     assert torch.equal(res[0]["params"], res[1]["params"])
+
+
+def test_acco_heterogeneous_accumulation_ws3():
+    """Three ranks, mixed accumulation (1/3/3 grads per round) over ragged
+    shards: bitwise rank agreement + exact oracle replay of the recorded
+    schedule."""
+    tmpdir = run_distributed(_worker_hetero, 3, timeout=300)
+    res = [torch.load(os.path.join(tmpdir, f"res_{r}.pt"),
+                      weights_only=False) for r in range(3)]
+    assert torch.equal(res[0]["params"], res[1]["params"])
+    assert torch.equal(res[0]["params"], res[2]["params"])
+    assert res[0]["count"] >= TARGET
+
+    torch.manual_seed(7)
+    w0 = nn.Linear(D, 1, bias=False).weight.detach().view(-1).clone()
+    traces = {r: res[r]["trace"] for r in range(3)}
+    P, count_tot = oracle_replay(w0, traces, 3, 0)
+    assert count_tot == res[0]["count"]
+    assert torch.allclose(P, res[0]["params"], atol=1e-6, rtol=1e-6)
