@@ -4,7 +4,6 @@ if it is not built — no silent eager fallback on a GPU box), and
 ACCO_FORCE_REF=1 is the only override. Covers the driver's
 "native code not loaded" failure mode from the CPU side."""
 
-import os
 import types
 
 import pytest

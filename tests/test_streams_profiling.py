@@ -2,7 +2,6 @@
 rocTX tracing hooks — these must be exact no-ops off-GPU so the ACCO
 state machine and trainers run identically under gloo tests."""
 
-import contextlib
 import subprocess
 import sys
 
