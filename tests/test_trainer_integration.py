@@ -248,3 +248,8 @@ def test_trainer_eval_cadence_ws2():
     res = torch.load(os.path.join(tmpdir, "ev_0.pt"), weights_only=False)
     assert res["n_evals"] >= 1          # cadence fired (rank 0 logs it)
     assert all(v == v and v > 0 for v in res["vals"])  # finite, positive
+
+
+def test_trainer_ddp_ws3_ragged():
+    tmpdir = run_distributed(_worker_ddp, 3, timeout=300)
+    _check(tmpdir, "ddp", world=3)
