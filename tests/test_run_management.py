@@ -50,3 +50,32 @@ def test_run_dir_and_multirun_end_to_end(tmp_path, monkeypatch):
         job_dir = os.path.dirname(r)
         assert glob.glob(os.path.join(job_dir, "scalars", "*", "*.jsonl"))
     assert os.getcwd() == str(tmp_path)     # chdir restored between jobs
+
+
+def test_main_cli_subprocess(tmp_path):
+    """The real user entry: `python main.py <overrides>` in a subprocess
+    (covers the argv/__main__ path the in-process tests bypass)."""
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, PYTHONPATH=repo, MASTER_ADDR="127.0.0.1")
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE"):
+        env.pop(k, None)       # earlier in-process tests leak these
+    import socket
+    with socket.socket() as s:   # parent's gloo store may hold the default
+        s.bind(("127.0.0.1", 0))
+        env["MASTER_PORT"] = str(s.getsockname()[1])
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "main.py"),
+         "train=dpu", "data=synthetic", "model=gptneo", "run_dir=.",
+         "model.hidden_size=32", "model.num_layers=1", "model.num_heads=2",
+         "model.vocab_size=64", "model.max_position_embeddings=32",
+         "train.nb_steps_tot=3", "train.batch_size=2",
+         "train.max_length=16", "train.use_mixed_precision=false",
+         "train.save=false", "train.dataloader_num_workers=0",
+         "train.dataloader_persistent_workers=false",
+         "data.n_train_sequences=8"],
+        capture_output=True, text=True, timeout=600, cwd=str(tmp_path),
+        env=env)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert os.path.exists(tmp_path / "results.csv"), r.stdout[-2000:]
