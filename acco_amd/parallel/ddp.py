@@ -30,6 +30,16 @@ embedding backward lands.
 This moves strictly fewer bytes than the reference baseline
 (reduce-scatter + all-gather vs all-reduce + shard broadcast) on the
 per-link-bound xGMI fabric.
+
+Cross-rank collective ordering (the RCCL/NCCL hazard): collectives on one
+communicator must be issued in the same order on every rank or the ranks
+can deadlock. Overlapped bucket launches are safe here because autograd
+uses ONE worker thread per device and every rank executes the identical
+backward graph, so producers finish — and coverage thresholds trip — in
+the same deterministic order on all ranks; the lock only guards the
+counter arithmetic, not ordering. Buckets never launched in backward
+(padding-only or no-grad buckets) are launched by ``finish_step`` in
+ascending bucket order, again identical on all ranks.
 """
 
 from __future__ import annotations
