@@ -84,3 +84,10 @@ def test_bench_refuses_mislabeled_gpus_flag():
         capture_output=True, text=True, timeout=300, cwd=REPO)
     assert out.returncode == 2
     assert not [l for l in out.stdout.splitlines() if l.startswith("{")]
+
+
+def test_bench_json_contract_ckpt_flag():
+    """--ckpt (activation checkpointing) runs and is recorded in config."""
+    d = _run(["--steps", "2", "--warmup", "1", "--ckpt"])
+    assert d["config"]["ckpt"] is True
+    assert d["value"] > 0
